@@ -1,0 +1,7 @@
+from .synthetic import (  # noqa: F401
+    SyntheticAN4,
+    SyntheticImages,
+    SyntheticPTB,
+    make_dataset,
+    make_loader,
+)

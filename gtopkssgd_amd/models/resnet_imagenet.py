@@ -1,0 +1,110 @@
+"""ImageNet ResNet-50 (bottleneck blocks) and AlexNet.
+
+Reference parity: the reference pulls these from torchvision
+(SURVEY.md §2.1 "ImageNet models"); torchvision is not in this
+environment, so they are implemented here from the papers.  ResNet-50
+is the headline benchmark model (BASELINE.json).
+"""
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+class Bottleneck(nn.Module):
+    expansion = 4
+
+    def __init__(self, in_planes, planes, stride=1, downsample=None):
+        super().__init__()
+        self.conv1 = nn.Conv2d(in_planes, planes, 1, bias=False)
+        self.bn1 = nn.BatchNorm2d(planes)
+        self.conv2 = nn.Conv2d(planes, planes, 3, stride=stride, padding=1,
+                               bias=False)
+        self.bn2 = nn.BatchNorm2d(planes)
+        self.conv3 = nn.Conv2d(planes, planes * 4, 1, bias=False)
+        self.bn3 = nn.BatchNorm2d(planes * 4)
+        self.downsample = downsample
+        self.stride = stride
+
+    def forward(self, x):
+        identity = x
+        out = F.relu(self.bn1(self.conv1(x)), inplace=True)
+        out = F.relu(self.bn2(self.conv2(out)), inplace=True)
+        out = self.bn3(self.conv3(out))
+        if self.downsample is not None:
+            identity = self.downsample(x)
+        return F.relu(out + identity, inplace=True)
+
+
+class ResNet(nn.Module):
+    def __init__(self, layers, num_classes=1000):
+        super().__init__()
+        self.in_planes = 64
+        self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
+        self.bn1 = nn.BatchNorm2d(64)
+        self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
+        self.layer1 = self._make_layer(64, layers[0])
+        self.layer2 = self._make_layer(128, layers[1], stride=2)
+        self.layer3 = self._make_layer(256, layers[2], stride=2)
+        self.layer4 = self._make_layer(512, layers[3], stride=2)
+        self.fc = nn.Linear(512 * 4, num_classes)
+        for m in self.modules():
+            if isinstance(m, nn.Conv2d):
+                nn.init.kaiming_normal_(m.weight, mode="fan_out",
+                                        nonlinearity="relu")
+            elif isinstance(m, nn.BatchNorm2d):
+                nn.init.ones_(m.weight)
+                nn.init.zeros_(m.bias)
+
+    def _make_layer(self, planes, blocks, stride=1):
+        downsample = None
+        if stride != 1 or self.in_planes != planes * 4:
+            downsample = nn.Sequential(
+                nn.Conv2d(self.in_planes, planes * 4, 1, stride=stride,
+                          bias=False),
+                nn.BatchNorm2d(planes * 4))
+        layers = [Bottleneck(self.in_planes, planes, stride, downsample)]
+        self.in_planes = planes * 4
+        for _ in range(1, blocks):
+            layers.append(Bottleneck(self.in_planes, planes))
+        return nn.Sequential(*layers)
+
+    def forward(self, x):
+        x = self.maxpool(F.relu(self.bn1(self.conv1(x)), inplace=True))
+        x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
+        x = F.adaptive_avg_pool2d(x, 1).flatten(1)
+        return self.fc(x)
+
+
+def resnet50(num_classes=1000):
+    return ResNet([3, 4, 6, 3], num_classes)
+
+
+def resnet101(num_classes=1000):
+    return ResNet([3, 4, 23, 3], num_classes)
+
+
+class AlexNet(nn.Module):
+    def __init__(self, num_classes=1000):
+        super().__init__()
+        self.features = nn.Sequential(
+            nn.Conv2d(3, 64, 11, stride=4, padding=2), nn.ReLU(inplace=True),
+            nn.MaxPool2d(3, 2),
+            nn.Conv2d(64, 192, 5, padding=2), nn.ReLU(inplace=True),
+            nn.MaxPool2d(3, 2),
+            nn.Conv2d(192, 384, 3, padding=1), nn.ReLU(inplace=True),
+            nn.Conv2d(384, 256, 3, padding=1), nn.ReLU(inplace=True),
+            nn.Conv2d(256, 256, 3, padding=1), nn.ReLU(inplace=True),
+            nn.MaxPool2d(3, 2))
+        self.classifier = nn.Sequential(
+            nn.Dropout(), nn.Linear(256 * 6 * 6, 4096), nn.ReLU(inplace=True),
+            nn.Dropout(), nn.Linear(4096, 4096), nn.ReLU(inplace=True),
+            nn.Linear(4096, num_classes))
+
+    def forward(self, x):
+        x = self.features(x)
+        x = torch.flatten(F.adaptive_avg_pool2d(x, 6), 1)
+        return self.classifier(x)
+
+
+def alexnet(num_classes=1000):
+    return AlexNet(num_classes)

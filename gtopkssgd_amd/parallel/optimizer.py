@@ -1,0 +1,284 @@
+"""Horovod-style DistributedOptimizer wrapper for gTop-k S-SGD.
+
+Reference parity: `upstream:distributed_optimizer.py`
+— wraps any ``torch.optim.Optimizer``; ``step()`` synchronises the
+globally-merged sparse gradient and then applies the update;
+``DistributedOptimizer(...)`` factory; ``rank()`` / ``size()`` helpers;
+initial parameter broadcast from rank 0 (C5).
+
+MI355X-native design differences (deliberate):
+
+* **Flat gradient buffer** — every ``param.grad`` is a VIEW into one
+  contiguous fp32 buffer, so backward accumulates the whole-model
+  gradient in place.  The reference re-flattens/concats per iteration
+  and stages through CPU numpy; here the flat vector already exists on
+  the GPU, sized for 288 GB HBM3E (one extra n-vector each for
+  residual, dense merge scratch, momentum).
+* **No background comm thread** — the merge is stream-ordered RCCL work
+  launched from ``step()``; a Python thread would serialise on the GIL
+  and buy nothing on xGMI (the reference needed one to hide 1 Gbps
+  Ethernet latency behind compute).
+* **Fused flat update** — when the inner optimizer is a single-group
+  ``torch.optim.SGD``, parameters and momentum also live in flat
+  buffers and the whole-model update is ONE HIP kernel
+  (:func:`gtopkssgd_amd.ops.sgd_step_`) instead of per-tensor launches.
+  Any other optimizer falls back to densify-into-``.grad`` + inner
+  ``step()`` — the reference's exact semantics.
+"""
+from __future__ import annotations
+
+from typing import Iterable, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+
+from .. import ops
+from ..compression import NoneCompressor, TopKCompressor, compressors
+from ..settings import WARMUP_ITERS, logger
+from .allreducer import GradReducer
+
+
+def rank() -> int:
+    if dist.is_available() and dist.is_initialized():
+        return dist.get_rank()
+    return 0
+
+
+def size() -> int:
+    if dist.is_available() and dist.is_initialized():
+        return dist.get_world_size()
+    return 1
+
+
+class _DistributedOptimizer(torch.optim.Optimizer):
+    def __init__(self, optimizer: torch.optim.Optimizer,
+                 named_parameters: Optional[Iterable[Tuple[str, torch.nn.Parameter]]] = None,
+                 compression=None,
+                 is_sparse: bool = True,
+                 density: float = 0.001,
+                 mode: Optional[str] = None,
+                 group=None,
+                 err_callback=None,
+                 warmup_iters: int = WARMUP_ITERS):
+        self.optimizer = optimizer
+        self.group = group
+        self.err_callback = err_callback
+        self.warmup_iters = warmup_iters
+        self.train_iter = 0
+        self._synchronized = False
+
+        if compression is None:
+            compression = (TopKCompressor(density) if is_sparse
+                           else NoneCompressor())
+        self.compression = compression
+        if mode is None:
+            if isinstance(compression, NoneCompressor) or not is_sparse:
+                mode = "dense"
+            else:
+                mode = getattr(compression, "name", "gtopk")
+                if mode not in ("gtopk", "topk"):
+                    mode = "gtopk"
+        assert mode in ("gtopk", "topk", "dense"), mode
+        self.mode = mode
+
+        if named_parameters is not None:
+            named_parameters = list(named_parameters)
+            self._params = [p for _, p in named_parameters if p.requires_grad]
+        else:
+            self._params = [p for g in optimizer.param_groups
+                            for p in g["params"] if p.requires_grad]
+        assert len(self._params) > 0
+        device = self._params[0].device
+        dtype = self._params[0].dtype
+        assert all(p.device == device and p.dtype == dtype
+                   for p in self._params), \
+            "all parameters must share one device/dtype"
+        assert dtype == torch.float32, \
+            "master parameters must be fp32 (bf16 compute via autocast)"
+        self.device = device
+        self.numel = sum(p.numel() for p in self._params)
+
+        # --- flat gradient buffer: p.grad are views ---
+        self.flat_grad = torch.zeros(self.numel, dtype=torch.float32,
+                                     device=device)
+        offset = 0
+        self._offsets = []
+        for p in self._params:
+            n = p.numel()
+            p.grad = self.flat_grad[offset:offset + n].view_as(p)
+            self._offsets.append(offset)
+            offset += n
+
+        # --- fused flat-SGD path ---
+        self._fused = self._can_fuse(optimizer)
+        if self._fused:
+            self.flat_param = torch.empty(self.numel, dtype=torch.float32,
+                                          device=device)
+            offset = 0
+            for p in self._params:
+                n = p.numel()
+                self.flat_param[offset:offset + n].copy_(p.data.view(-1))
+                p.data = self.flat_param[offset:offset + n].view_as(p)
+                offset += n
+            self.flat_momentum = torch.zeros_like(self.flat_param)
+        else:
+            self.flat_param = None
+            self.flat_momentum = None
+
+        density_eff = getattr(compression, "density", density)
+        self.k = max(1, int(density_eff * self.numel)) \
+            if self.mode != "dense" else self.numel
+        self.reducer = GradReducer(self.numel, self.k, device,
+                                   mode=self.mode, group=group)
+
+        self._broadcast_parameters()
+        logger.info(
+            "DistributedOptimizer: n=%d k=%d mode=%s fused=%s world=%d",
+            self.numel, self.k, self.mode, self._fused, size())
+
+    # `torch.optim.Optimizer.__init__` is intentionally NOT called; we
+    # delegate param_groups/state to the inner optimizer.
+
+    @staticmethod
+    def _can_fuse(optimizer: torch.optim.Optimizer) -> bool:
+        if type(optimizer) is not torch.optim.SGD:
+            return False
+        if len(optimizer.param_groups) != 1:
+            return False
+        g = optimizer.param_groups[0]
+        # dampening breaks first-step parity with torch's clone-init
+        if g.get("dampening", 0.0) != 0.0:
+            return False
+        if g.get("maximize", False):
+            return False
+        return True
+
+    # ------------------------------------------------------------------
+    @property
+    def param_groups(self):
+        return self.optimizer.param_groups
+
+    @param_groups.setter
+    def param_groups(self, value):  # pragma: no cover - setter for compat
+        self.optimizer.param_groups = value
+
+    @property
+    def state(self):
+        return self.optimizer.state
+
+    @property
+    def defaults(self):
+        return self.optimizer.defaults
+
+    # ------------------------------------------------------------------
+    def _broadcast_parameters(self):
+        if size() == 1:
+            return
+        if self._fused:
+            dist.broadcast(self.flat_param, src=0, group=self.group)
+        else:
+            for p in self._params:
+                dist.broadcast(p.data, src=0, group=self.group)
+
+    def zero_grad(self, set_to_none: bool = False):
+        # views cannot be set to None — always zero the flat buffer
+        self.flat_grad.zero_()
+
+    # ------------------------------------------------------------------
+    def synchronize(self):
+        """Run compression + the global merge; leaves the merged gradient
+        staged for the update.  Idempotent per iteration."""
+        if self._synchronized:
+            return
+        world = self.reducer.world_size()
+        warmup = self.train_iter < self.warmup_iters
+        if self.mode == "dense" or warmup:
+            self.reducer.dense(self.flat_grad)  # averaged in place
+            self._staged = ("dense", None, None)
+        elif self.mode == "gtopk":
+            vals, idx = self.compression.compress(self.flat_grad, "flat")
+            gval, gidx = self.reducer.gtopk(vals, idx)
+            self.compression.add_residuals(gidx, "flat")
+            self._staged = ("sparse", gval, gidx)
+        else:  # topk baseline: union sum sits in dense_scratch
+            vals, idx = self.compression.compress(self.flat_grad, "flat")
+            self.reducer.topk(vals, idx)
+            # every local pick is included in the union → nothing rejected
+            self._staged = ("union", None, None)
+        self._synchronized = True
+
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        self.synchronize()
+        kind, gval, gidx = self._staged
+        world = self.reducer.world_size()
+        g = self.optimizer.param_groups[0]
+        if self._fused:
+            if kind == "dense":
+                grad_buf, scale, zero = self.flat_grad, 1.0, True
+            elif kind == "sparse":
+                ops.densify_(self.reducer.dense_scratch, gidx, gval, 1.0)
+                grad_buf, scale, zero = (self.reducer.dense_scratch,
+                                         1.0 / world, True)
+                self.flat_grad.zero_()
+            else:  # union
+                grad_buf, scale, zero = (self.reducer.dense_scratch,
+                                         1.0 / world, True)
+                self.flat_grad.zero_()
+            ops.sgd_step_(self.flat_param, self.flat_momentum, grad_buf,
+                          lr=g["lr"], momentum=g.get("momentum", 0.0),
+                          weight_decay=g.get("weight_decay", 0.0),
+                          dampening=0.0, nesterov=g.get("nesterov", False),
+                          zero_grad=zero, grad_scale=scale)
+        else:
+            if kind == "sparse":
+                self.flat_grad.zero_()
+                ops.densify_(self.flat_grad, gidx, gval, 1.0 / world)
+            elif kind == "union":
+                self.flat_grad.zero_()
+                self.flat_grad.add_(self.reducer.dense_scratch,
+                                    alpha=1.0 / world)
+                self.reducer.dense_scratch.zero_()
+            self.optimizer.step()
+            if kind != "dense":
+                self.flat_grad.zero_()
+        self.train_iter += 1
+        self._synchronized = False
+        return loss
+
+    # ------------------------------------------------------------------
+    def state_dict(self):
+        sd = {"inner": self.optimizer.state_dict(),
+              "train_iter": self.train_iter,
+              "compression": self.compression.state_dict()}
+        if self._fused:
+            sd["flat_momentum"] = self.flat_momentum
+        return sd
+
+    def load_state_dict(self, sd):
+        self.optimizer.load_state_dict(sd["inner"])
+        self.train_iter = sd.get("train_iter", 0)
+        self.compression.load_state_dict(sd.get("compression", {}))
+        if self._fused and "flat_momentum" in sd:
+            self.flat_momentum.copy_(sd["flat_momentum"].to(self.device))
+
+    def add_param_group(self, group):  # pragma: no cover - compat
+        raise NotImplementedError(
+            "DistributedOptimizer flattens parameters at construction; "
+            "add_param_group afterwards is unsupported")
+
+
+def DistributedOptimizer(optimizer, named_parameters=None, compression=None,
+                         is_sparse: bool = True, density: float = 0.001,
+                         mode: Optional[str] = None, group=None,
+                         err_callback=None,
+                         warmup_iters: int = WARMUP_ITERS):
+    """Factory mirroring the reference's
+    ``DistributedOptimizer(optimizer, named_parameters, compression,
+    is_sparse, density, err_callback, ...)``."""
+    return _DistributedOptimizer(optimizer, named_parameters, compression,
+                                 is_sparse, density, mode, group,
+                                 err_callback, warmup_iters)
