@@ -1,0 +1,441 @@
+// gtopkssgd_amd._C — hand-written HIP/CDNA4 (gfx950) kernels for the
+// gTop-k S-SGD hot path.  SURVEY.md §2.2 K1–K8.
+//
+// Design notes (MI355X-first, see /opt/skills/guides/cdna_hip_programming.md):
+//  * wave64 everywhere; __ballot is 64-bit; blocks are 256 threads.
+//  * All streaming kernels are grid-stride, float4-vectorized, capped at
+//    2048 blocks (Guideline 11).
+//  * Top-k select is a bucketed threshold select: one fused pass computes
+//    `residual += grad` AND an 8192-bin LDS histogram of the |value|
+//    bit-key (bits 30..18: exponent + top-5 mantissa); a single-block
+//    scan finds the threshold bucket; a second pass compacts winners with
+//    wave-aggregated atomics and zeroes them in the residual (error
+//    feedback).  Exact in count (always exactly k), ties inside the
+//    threshold bucket (a ≤3% relative-|v| band) are arbitrary — the
+//    reference's torch.topk is tie-arbitrary too.
+//  * fp32 global atomicAdd/atomicExch are native on CDNA4; scatter sizes
+//    here are O(k) ≈ 25k elements, contention-free in practice
+//    (Guideline 12: per-wave aggregation for the hot counters).
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <hip/hip_runtime.h>
+
+#define HIST_BITS 13
+#define HIST_BINS (1 << HIST_BITS)   // 8192
+#define KEY_SHIFT (31 - HIST_BITS)   // 18
+#define BLOCK 256
+#define MAX_BLOCKS 2048
+
+static inline int stream_blocks(long long work_items) {
+    long long b = (work_items + BLOCK - 1) / BLOCK;
+    if (b > MAX_BLOCKS) b = MAX_BLOCKS;
+    if (b < 1) b = 1;
+    return (int)b;
+}
+
+__device__ __forceinline__ unsigned mag_key(float v) {
+    unsigned bits = __float_as_uint(v) & 0x7fffffffu;
+    return bits >> KEY_SHIFT;
+}
+
+// ---------------------------------------------------------------------------
+// K1+K2 pass 1: residual += grad, histogram of |residual| keys
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(BLOCK) void ef_accum_hist_kernel(
+        float* __restrict__ residual, const float* __restrict__ grad,
+        long long n, unsigned* __restrict__ ghist) {
+    __shared__ unsigned lhist[HIST_BINS];   // 32 KiB LDS
+    for (int j = threadIdx.x; j < HIST_BINS; j += BLOCK) lhist[j] = 0;
+    __syncthreads();
+
+    const long long stride = (long long)gridDim.x * BLOCK;
+    const long long tid = (long long)blockIdx.x * BLOCK + threadIdx.x;
+    const long long n4 = n >> 2;
+    float4* r4 = reinterpret_cast<float4*>(residual);
+    const float4* g4 = reinterpret_cast<const float4*>(grad);
+    for (long long i = tid; i < n4; i += stride) {
+        float4 r = r4[i];
+        const float4 g = g4[i];
+        r.x += g.x; r.y += g.y; r.z += g.z; r.w += g.w;
+        r4[i] = r;
+        atomicAdd(&lhist[mag_key(r.x)], 1u);
+        atomicAdd(&lhist[mag_key(r.y)], 1u);
+        atomicAdd(&lhist[mag_key(r.z)], 1u);
+        atomicAdd(&lhist[mag_key(r.w)], 1u);
+    }
+    for (long long i = (n4 << 2) + tid; i < n; i += stride) {
+        float r = residual[i] + grad[i];
+        residual[i] = r;
+        atomicAdd(&lhist[mag_key(r)], 1u);
+    }
+    __syncthreads();
+    for (int j = threadIdx.x; j < HIST_BINS; j += BLOCK) {
+        unsigned c = lhist[j];
+        if (c) atomicAdd(&ghist[j], c);
+    }
+}
+
+// histogram-only variant for topk_among (candidate values)
+__global__ __launch_bounds__(BLOCK) void hist_vals_kernel(
+        const float* __restrict__ vals, long long n,
+        unsigned* __restrict__ ghist) {
+    __shared__ unsigned lhist[HIST_BINS];
+    for (int j = threadIdx.x; j < HIST_BINS; j += BLOCK) lhist[j] = 0;
+    __syncthreads();
+    const long long stride = (long long)gridDim.x * BLOCK;
+    for (long long i = (long long)blockIdx.x * BLOCK + threadIdx.x; i < n;
+         i += stride)
+        atomicAdd(&lhist[mag_key(vals[i])], 1u);
+    __syncthreads();
+    for (int j = threadIdx.x; j < HIST_BINS; j += BLOCK) {
+        unsigned c = lhist[j];
+        if (c) atomicAdd(&ghist[j], c);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// pass 2 (single block): find threshold bin T with
+//   count(bins > T) < k <= count(bins >= T); zero the compact counters.
+// info = [T, count_above]
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(BLOCK) void scan_threshold_kernel(
+        const unsigned* __restrict__ hist, int k, int* __restrict__ info,
+        int* __restrict__ counters) {
+    __shared__ unsigned chunk_prefix[BLOCK];
+    const int t = threadIdx.x;
+    const int chunk = HIST_BINS / BLOCK;        // 32 bins per thread
+    const int base_hi = HIST_BINS - 1 - t * chunk;
+    unsigned sum = 0;
+    for (int j = 0; j < chunk; ++j) sum += hist[base_hi - j];
+    chunk_prefix[t] = sum;
+    __syncthreads();
+    if (t == 0) {                               // serial 256-add scan
+        unsigned acc = 0;
+        for (int c = 0; c < BLOCK; ++c) {
+            unsigned s = chunk_prefix[c];
+            chunk_prefix[c] = acc;
+            acc += s;
+        }
+        counters[0] = 0;
+        counters[1] = 0;
+    }
+    __syncthreads();
+    unsigned cum = chunk_prefix[t];             // count strictly above chunk
+    for (int j = 0; j < chunk; ++j) {
+        const unsigned h = hist[base_hi - j];
+        if (cum < (unsigned)k && cum + h >= (unsigned)k) {
+            info[0] = base_hi - j;              // unique writer
+            info[1] = (int)cum;
+        }
+        cum += h;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// pass 3: compact winners.  Wave-aggregated slot allocation; winners in
+// the strict-above region always land; threshold-bin ties fill the rest.
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ int wave_alloc(int* counter, bool pred,
+                                          unsigned long long* out_mask) {
+    // returns this lane's rank-offset slot base via one atomic per wave;
+    // caller adds its own rank.  -1 when !pred.
+    const unsigned long long mask = __ballot(pred);
+    *out_mask = mask;
+    if (!pred) return -1;
+    const int lane = threadIdx.x & 63;
+    const unsigned long long lt = (lane == 63)
+        ? 0x7fffffffffffffffull : ((1ull << lane) - 1ull);
+    const int rank = __popcll(mask & lt);
+    int base = 0;
+    const int leader = __ffsll((unsigned long long)mask) - 1;
+    if (lane == leader) base = atomicAdd(counter, __popcll(mask));
+    base = __shfl(base, leader);
+    return base + rank;
+}
+
+template <bool ZERO_SRC, bool HAS_IDX>
+__global__ __launch_bounds__(BLOCK) void compact_kernel(
+        float* __restrict__ src, const int* __restrict__ src_idx,
+        long long n, const int* __restrict__ info, int* __restrict__ counters,
+        float* __restrict__ out_val, int* __restrict__ out_idx, int k) {
+    const int T = info[0];
+    const int above = info[1];
+    const long long stride = (long long)gridDim.x * BLOCK;
+    for (long long i = (long long)blockIdx.x * BLOCK + threadIdx.x; i < n;
+         i += stride) {
+        const float v = src[i];
+        const int key = (int)mag_key(v);
+        unsigned long long m;
+        // strict-above winners: guaranteed slots [0, above)
+        int slot = wave_alloc(&counters[0], key > T, &m);
+        if (key > T) {
+            out_val[slot] = v;
+            out_idx[slot] = HAS_IDX ? src_idx[i] : (int)i;
+            if (ZERO_SRC) src[i] = 0.0f;
+        }
+        // threshold-bin ties: slots [above, k)
+        slot = wave_alloc(&counters[1], key == T, &m);
+        if (key == T) {
+            slot += above;
+            if (slot < k) {
+                out_val[slot] = v;
+                out_idx[slot] = HAS_IDX ? src_idx[i] : (int)i;
+                if (ZERO_SRC) src[i] = 0.0f;
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// K4: scatter-add / gather-zero (merge round primitives)
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(BLOCK) void scatter_add_kernel(
+        float* __restrict__ dense, const int* __restrict__ idx,
+        const float* __restrict__ val, long long m) {
+    const long long stride = (long long)gridDim.x * BLOCK;
+    for (long long i = (long long)blockIdx.x * BLOCK + threadIdx.x; i < m;
+         i += stride)
+        atomicAdd(&dense[idx[i]], val[i]);
+}
+
+__global__ __launch_bounds__(BLOCK) void gather_zero_kernel(
+        float* __restrict__ dense, const int* __restrict__ idx,
+        float* __restrict__ out, long long m) {
+    const long long stride = (long long)gridDim.x * BLOCK;
+    for (long long i = (long long)blockIdx.x * BLOCK + threadIdx.x; i < m;
+         i += stride)
+        out[i] = atomicExch(&dense[idx[i]], 0.0f);  // dedup: dup reads 0
+}
+
+// ---------------------------------------------------------------------------
+// K6: rejected-value re-insertion (marks: byte per element, reused)
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(BLOCK) void mark_set_kernel(
+        unsigned char* __restrict__ marks, const int* __restrict__ idx,
+        long long m, unsigned char value) {
+    const long long stride = (long long)gridDim.x * BLOCK;
+    for (long long i = (long long)blockIdx.x * BLOCK + threadIdx.x; i < m;
+         i += stride)
+        marks[idx[i]] = value;
+}
+
+__global__ __launch_bounds__(BLOCK) void readd_kernel(
+        float* __restrict__ residual, const int* __restrict__ lidx,
+        const float* __restrict__ lval, long long m,
+        const unsigned char* __restrict__ marks) {
+    const long long stride = (long long)gridDim.x * BLOCK;
+    for (long long i = (long long)blockIdx.x * BLOCK + threadIdx.x; i < m;
+         i += stride) {
+        const int j = lidx[i];
+        if (!marks[j]) residual[j] += lval[i];  // lidx unique: no atomics
+    }
+}
+
+// ---------------------------------------------------------------------------
+// densify: dense[idx] = val * scale, skipping zeros (dedup artifacts)
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(BLOCK) void densify_kernel(
+        float* __restrict__ dense, const int* __restrict__ idx,
+        const float* __restrict__ val, long long m, float scale) {
+    const long long stride = (long long)gridDim.x * BLOCK;
+    for (long long i = (long long)blockIdx.x * BLOCK + threadIdx.x; i < m;
+         i += stride) {
+        const float v = val[i];
+        if (v != 0.0f) dense[idx[i]] = v * scale;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// K7: fused flat SGD (momentum + weight decay + optional nesterov),
+// reading the dense grad scratch and zeroing it in the same pass.
+// Memory-bound: float4-vectorized, ~24 B/element of traffic.
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ float sgd_one(float p, float& mbuf, float g,
+                                         float lr, float mu, float wd,
+                                         float damp, bool nesterov,
+                                         float gscale) {
+    float gg = g * gscale + wd * p;
+    if (mu != 0.0f) {
+        mbuf = mu * mbuf + (1.0f - damp) * gg;
+        gg = nesterov ? (gg + mu * mbuf) : mbuf;
+    }
+    return p - lr * gg;
+}
+
+__global__ __launch_bounds__(BLOCK) void sgd_kernel(
+        float* __restrict__ param, float* __restrict__ mom,
+        float* __restrict__ grad, long long n, float lr, float mu, float wd,
+        float damp, int nesterov, int zero_grad, float gscale) {
+    const long long stride = (long long)gridDim.x * BLOCK;
+    const long long tid = (long long)blockIdx.x * BLOCK + threadIdx.x;
+    const long long n4 = n >> 2;
+    float4* p4 = reinterpret_cast<float4*>(param);
+    float4* m4 = reinterpret_cast<float4*>(mom);
+    float4* g4 = reinterpret_cast<float4*>(grad);
+    const bool nest = nesterov != 0;
+    for (long long i = tid; i < n4; i += stride) {
+        float4 p = p4[i], m = m4[i];
+        const float4 g = g4[i];
+        p.x = sgd_one(p.x, m.x, g.x, lr, mu, wd, damp, nest, gscale);
+        p.y = sgd_one(p.y, m.y, g.y, lr, mu, wd, damp, nest, gscale);
+        p.z = sgd_one(p.z, m.z, g.z, lr, mu, wd, damp, nest, gscale);
+        p.w = sgd_one(p.w, m.w, g.w, lr, mu, wd, damp, nest, gscale);
+        p4[i] = p;
+        m4[i] = m;
+        if (zero_grad) g4[i] = make_float4(0.f, 0.f, 0.f, 0.f);
+    }
+    for (long long i = (n4 << 2) + tid; i < n; i += stride) {
+        float m = mom[i];
+        param[i] = sgd_one(param[i], m, grad[i], lr, mu, wd, damp, nest,
+                           gscale);
+        mom[i] = m;
+        if (zero_grad) grad[i] = 0.0f;
+    }
+}
+
+// ===========================================================================
+// Host wrappers
+// ===========================================================================
+#define CHECK_F32(t) TORCH_CHECK((t).is_cuda() && (t).is_contiguous() && \
+    (t).scalar_type() == at::kFloat, #t " must be contiguous fp32 CUDA")
+#define CHECK_I32(t) TORCH_CHECK((t).is_cuda() && (t).is_contiguous() && \
+    (t).scalar_type() == at::kInt, #t " must be contiguous int32 CUDA")
+
+static hipStream_t cur_stream() {
+    return at::cuda::getCurrentCUDAStream().stream();
+}
+
+#define HIP_OK(expr) do { hipError_t _e = (expr); \
+    TORCH_CHECK(_e == hipSuccess, "HIP error: ", hipGetErrorString(_e)); \
+} while (0)
+
+static void check_launch(const char* what) {
+    HIP_OK(hipGetLastError());
+    (void)what;
+}
+
+void ef_compress(at::Tensor residual, at::Tensor grad, int64_t k,
+                 at::Tensor hist, at::Tensor info, at::Tensor counters,
+                 at::Tensor out_val, at::Tensor out_idx) {
+    CHECK_F32(residual); CHECK_F32(grad); CHECK_F32(out_val);
+    CHECK_I32(hist); CHECK_I32(info); CHECK_I32(counters); CHECK_I32(out_idx);
+    const long long n = residual.numel();
+    TORCH_CHECK(grad.numel() == n && k > 0 && k <= n);
+    TORCH_CHECK(hist.numel() >= HIST_BINS && out_val.numel() >= k &&
+                out_idx.numel() >= k);
+    auto stream = cur_stream();
+    HIP_OK(hipMemsetAsync(hist.data_ptr<int>(), 0, HIST_BINS * sizeof(int), stream));
+    const int blocks = stream_blocks((n + 3) / 4);
+    ef_accum_hist_kernel<<<blocks, BLOCK, 0, stream>>>(
+        residual.data_ptr<float>(), grad.data_ptr<float>(), n,
+        (unsigned*)hist.data_ptr<int>());
+    scan_threshold_kernel<<<1, BLOCK, 0, stream>>>(
+        (unsigned*)hist.data_ptr<int>(), (int)k, info.data_ptr<int>(),
+        counters.data_ptr<int>());
+    compact_kernel<true, false><<<stream_blocks(n), BLOCK, 0, stream>>>(
+        residual.data_ptr<float>(), nullptr, n, info.data_ptr<int>(),
+        counters.data_ptr<int>(), out_val.data_ptr<float>(),
+        out_idx.data_ptr<int>(), (int)k);
+    check_launch("gtopk");
+}
+
+void topk_among(at::Tensor values, at::Tensor indexes, int64_t k,
+                at::Tensor hist, at::Tensor info, at::Tensor counters,
+                at::Tensor out_val, at::Tensor out_idx) {
+    CHECK_F32(values); CHECK_F32(out_val);
+    CHECK_I32(indexes); CHECK_I32(hist); CHECK_I32(info);
+    CHECK_I32(counters); CHECK_I32(out_idx);
+    const long long m = values.numel();
+    TORCH_CHECK(indexes.numel() == m && k > 0 && k <= m);
+    auto stream = cur_stream();
+    HIP_OK(hipMemsetAsync(hist.data_ptr<int>(), 0, HIST_BINS * sizeof(int), stream));
+    const int blocks = stream_blocks(m);
+    hist_vals_kernel<<<blocks, BLOCK, 0, stream>>>(
+        values.data_ptr<float>(), m, (unsigned*)hist.data_ptr<int>());
+    scan_threshold_kernel<<<1, BLOCK, 0, stream>>>(
+        (unsigned*)hist.data_ptr<int>(), (int)k, info.data_ptr<int>(),
+        counters.data_ptr<int>());
+    compact_kernel<false, true><<<blocks, BLOCK, 0, stream>>>(
+        values.data_ptr<float>(), indexes.data_ptr<int>(), m,
+        info.data_ptr<int>(), counters.data_ptr<int>(),
+        out_val.data_ptr<float>(), out_idx.data_ptr<int>(), (int)k);
+    check_launch("gtopk");
+}
+
+void scatter_add(at::Tensor dense, at::Tensor idx, at::Tensor val) {
+    CHECK_F32(dense); CHECK_I32(idx); CHECK_F32(val);
+    const long long m = idx.numel();
+    TORCH_CHECK(val.numel() == m);
+    scatter_add_kernel<<<stream_blocks(m), BLOCK, 0, cur_stream()>>>(
+        dense.data_ptr<float>(), idx.data_ptr<int>(), val.data_ptr<float>(),
+        m);
+    check_launch("gtopk");
+}
+
+void gather_zero(at::Tensor dense, at::Tensor idx, at::Tensor out) {
+    CHECK_F32(dense); CHECK_I32(idx); CHECK_F32(out);
+    const long long m = idx.numel();
+    TORCH_CHECK(out.numel() == m);
+    gather_zero_kernel<<<stream_blocks(m), BLOCK, 0, cur_stream()>>>(
+        dense.data_ptr<float>(), idx.data_ptr<int>(), out.data_ptr<float>(),
+        m);
+    check_launch("gtopk");
+}
+
+void readd_rejected(at::Tensor residual, at::Tensor lidx, at::Tensor lval,
+                    at::Tensor gidx, at::Tensor marks) {
+    CHECK_F32(residual); CHECK_I32(lidx); CHECK_F32(lval); CHECK_I32(gidx);
+    TORCH_CHECK(marks.is_cuda() && marks.scalar_type() == at::kByte &&
+                marks.numel() == residual.numel());
+    auto stream = cur_stream();
+    const long long mg = gidx.numel(), ml = lidx.numel();
+    TORCH_CHECK(lval.numel() == ml);
+    auto* mk = marks.data_ptr<unsigned char>();
+    mark_set_kernel<<<stream_blocks(mg), BLOCK, 0, stream>>>(
+        mk, gidx.data_ptr<int>(), mg, 1);
+    readd_kernel<<<stream_blocks(ml), BLOCK, 0, stream>>>(
+        residual.data_ptr<float>(), lidx.data_ptr<int>(),
+        lval.data_ptr<float>(), ml, mk);
+    mark_set_kernel<<<stream_blocks(mg), BLOCK, 0, stream>>>(
+        mk, gidx.data_ptr<int>(), mg, 0);
+    check_launch("gtopk");
+}
+
+void densify(at::Tensor dense, at::Tensor idx, at::Tensor val, double scale) {
+    CHECK_F32(dense); CHECK_I32(idx); CHECK_F32(val);
+    const long long m = idx.numel();
+    TORCH_CHECK(val.numel() == m);
+    densify_kernel<<<stream_blocks(m), BLOCK, 0, cur_stream()>>>(
+        dense.data_ptr<float>(), idx.data_ptr<int>(), val.data_ptr<float>(),
+        m, (float)scale);
+    check_launch("gtopk");
+}
+
+void sgd_step(at::Tensor param, at::Tensor mom, at::Tensor grad, double lr,
+              double momentum, double weight_decay, double dampening,
+              bool nesterov, bool zero_grad, double grad_scale) {
+    CHECK_F32(param); CHECK_F32(mom); CHECK_F32(grad);
+    const long long n = param.numel();
+    TORCH_CHECK(mom.numel() == n && grad.numel() == n);
+    sgd_kernel<<<stream_blocks((n + 3) / 4), BLOCK, 0, cur_stream()>>>(
+        param.data_ptr<float>(), mom.data_ptr<float>(),
+        grad.data_ptr<float>(), n, (float)lr, (float)momentum,
+        (float)weight_decay, (float)dampening, nesterov ? 1 : 0,
+        zero_grad ? 1 : 0, (float)grad_scale);
+    check_launch("gtopk");
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("ef_compress", &ef_compress,
+          "fused residual accumulate + bucketed top-k select (gfx950)");
+    m.def("topk_among", &topk_among, "top-k among candidate (val,idx) pairs");
+    m.def("scatter_add", &scatter_add, "dense[idx] += val");
+    m.def("gather_zero", &gather_zero, "out = dense[idx]; dense[idx] = 0");
+    m.def("readd_rejected", &readd_rejected,
+          "residual[lidx] += lval where lidx not in gidx");
+    m.def("densify", &densify, "dense[idx] = val * scale (skip zeros)");
+    m.def("sgd_step", &sgd_step,
+          "fused flat SGD w/ momentum+wd+nesterov, self-zeroing grad");
+    m.attr("HIST_BINS") = HIST_BINS;
+}
