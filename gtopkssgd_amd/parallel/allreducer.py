@@ -194,3 +194,84 @@ class GradReducer:
 
     def dense(self, flat_grad):
         return dense_allreduce(flat_grad, self.group)
+
+
+def _self_benchmark():  # pragma: no cover - exercised on GPU boxes
+    """Collective self-benchmark (reference parity: the ``__main__``
+    block of `upstream:allreducer.py` benchmarks gtopk_sparse_allreduce
+    on random vectors under mpirun).
+
+        python -m torch.distributed.run --nproc-per-node N \
+            --master-addr 127.0.0.1 -m gtopkssgd_amd.parallel.allreducer \
+            --n 25557032 --density 0.001 --iters 50
+    """
+    import argparse
+    import json
+    import os
+    import time
+
+    p = argparse.ArgumentParser()
+    p.add_argument("--n", type=int, default=25_557_032)  # ResNet-50
+    p.add_argument("--density", type=float, default=0.001)
+    p.add_argument("--iters", type=int, default=50)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--mode", default="gtopk",
+                   choices=["gtopk", "topk", "dense"])
+    p.add_argument("--device", default=None)
+    args = p.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    use_gpu = torch.cuda.is_available() and args.device != "cpu"
+    if world > 1:
+        dist.init_process_group("nccl" if use_gpu else "gloo")
+        if use_gpu:
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+    device = torch.device("cuda" if use_gpu else "cpu")
+
+    from ..compression import TopKCompressor
+    n = args.n
+    k = max(1, int(args.density * n))
+    comp = TopKCompressor(args.density)
+    comp.residuals["b"] = torch.zeros(n, device=device)
+    reducer = GradReducer(n, k, device, mode=args.mode)
+    g = torch.Generator(device="cpu").manual_seed(rank)
+    grad = torch.randn(n, generator=g).to(device)
+
+    def one_iter():
+        if args.mode == "dense":
+            reducer.dense(grad.clone())
+            return
+        vals, idx = comp.compress(grad, "b")
+        if args.mode == "gtopk":
+            gval, gidx = reducer.gtopk(vals, idx)
+            comp.add_residuals(gidx, "b")
+        else:
+            reducer.topk(vals, idx)
+            reducer.dense_scratch.zero_()
+
+    for _ in range(args.warmup):
+        one_iter()
+    if use_gpu:
+        torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    t0 = time.perf_counter()
+    for _ in range(args.iters):
+        one_iter()
+    if use_gpu:
+        torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    dt = (time.perf_counter() - t0) / args.iters
+    if rank == 0:
+        print(json.dumps({
+            "collective": args.mode, "n": n, "k": k, "world": world,
+            "ms_per_iter": round(dt * 1000, 3),
+            "device": str(device)}), flush=True)
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":  # pragma: no cover
+    _self_benchmark()

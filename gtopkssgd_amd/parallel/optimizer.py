@@ -99,15 +99,21 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         self.numel = sum(p.numel() for p in self._params)
 
         # --- flat gradient buffer: p.grad are views ---
+        # Views preserve each param's memory format (e.g. channels_last
+        # conv weights for MIOpen) by re-striding the contiguous chunk.
+        def _flat_view(flat: torch.Tensor, offset: int,
+                       p: torch.Tensor) -> torch.Tensor:
+            return flat[offset:offset + p.numel()].as_strided(
+                p.size(), p.stride())
+
         self.flat_grad = torch.zeros(self.numel, dtype=torch.float32,
                                      device=device)
         offset = 0
         self._offsets = []
         for p in self._params:
-            n = p.numel()
-            p.grad = self.flat_grad[offset:offset + n].view_as(p)
+            p.grad = _flat_view(self.flat_grad, offset, p)
             self._offsets.append(offset)
-            offset += n
+            offset += p.numel()
 
         # --- fused flat-SGD path ---
         self._fused = self._can_fuse(optimizer)
@@ -115,11 +121,12 @@ class _DistributedOptimizer(torch.optim.Optimizer):
             self.flat_param = torch.empty(self.numel, dtype=torch.float32,
                                           device=device)
             offset = 0
-            for p in self._params:
-                n = p.numel()
-                self.flat_param[offset:offset + n].copy_(p.data.view(-1))
-                p.data = self.flat_param[offset:offset + n].view_as(p)
-                offset += n
+            with torch.no_grad():
+                for p in self._params:
+                    view = _flat_view(self.flat_param, offset, p)
+                    view.copy_(p.data)
+                    p.data = view
+                    offset += p.numel()
             self.flat_momentum = torch.zeros_like(self.flat_param)
         else:
             self.flat_param = None

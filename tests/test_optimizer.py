@@ -154,3 +154,27 @@ def test_warmup_dense_iters():
         loss.backward()
         opt.step()
     assert opt.train_iter == 3
+
+
+def test_channels_last_params_preserved():
+    """Flat views must preserve channels_last conv-weight layout
+    (MIOpen wants channels_last; .view would have failed/broken it)."""
+    import torch.nn.functional as F
+    torch.manual_seed(0)
+    net = nn.Sequential(nn.Conv2d(3, 8, 3, padding=1), nn.ReLU(),
+                        nn.Conv2d(8, 4, 3, padding=1))
+    net = net.to(memory_format=torch.channels_last)
+    w0 = net[0].weight.detach().clone()
+    inner = torch.optim.SGD(net.parameters(), lr=0.1, momentum=0.9)
+    opt = gt.DistributedOptimizer(inner, net.named_parameters(),
+                                  compression=gt.compressors["gtopk"](1.0))
+    assert net[0].weight.is_contiguous(memory_format=torch.channels_last)
+    assert torch.equal(net[0].weight, w0)
+    x = torch.randn(2, 3, 16, 16).contiguous(
+        memory_format=torch.channels_last)
+    loss = net(x).square().mean()
+    loss.backward()
+    assert net[0].weight.grad.is_contiguous(
+        memory_format=torch.channels_last)
+    opt.step()
+    assert not torch.equal(net[0].weight, w0)  # update applied
