@@ -15,8 +15,6 @@ import json
 import os
 import time
 
-os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
-
 import torch  # noqa: E402
 import torch.distributed as dist  # noqa: E402
 import torch.nn as nn  # noqa: E402
@@ -51,6 +49,9 @@ def main():
         torch.cuda.set_device(local_rank)
     device = torch.device("cuda", local_rank)
     torch.manual_seed(1234)
+    # MIOpen: exhaustive-find conv algo selection (amortised in warmup);
+    # without it MIOpen falls back to slow zero-workspace solvers.
+    torch.backends.cudnn.benchmark = True
 
     model = resnet50().to(device).to(memory_format=torch.channels_last)
     inner = torch.optim.SGD(model.parameters(), lr=args.lr, momentum=0.9,

@@ -5,3 +5,4 @@ from .synthetic import (  # noqa: F401
     make_dataset,
     make_loader,
 )
+from .packed import PackedTensorDataset, pack_dataset  # noqa: F401

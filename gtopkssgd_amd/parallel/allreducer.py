@@ -216,7 +216,7 @@ def _self_benchmark():  # pragma: no cover - exercised on GPU boxes
     p.add_argument("--iters", type=int, default=50)
     p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--mode", default="gtopk",
-                   choices=["gtopk", "topk", "dense"])
+                   choices=["gtopk", "topk", "dense", "torchtopk"])
     p.add_argument("--device", default=None)
     args = p.parse_args()
 
@@ -241,6 +241,14 @@ def _self_benchmark():  # pragma: no cover - exercised on GPU boxes
     def one_iter():
         if args.mode == "dense":
             reducer.dense(grad.clone())
+            return
+        if args.mode == "torchtopk":
+            # the reference's select path: torch.topk over the flat vec
+            res = comp.residuals["b"]
+            res.add_(grad)
+            _, idx = torch.topk(res.abs(), k, sorted=False)
+            vals = res[idx]
+            res[idx] = 0
             return
         vals, idx = comp.compress(grad, "b")
         if args.mode == "gtopk":

@@ -59,13 +59,18 @@ class _DistributedOptimizer(torch.optim.Optimizer):
                  mode: Optional[str] = None,
                  group=None,
                  err_callback=None,
-                 warmup_iters: int = WARMUP_ITERS):
+                 warmup_iters: int = WARMUP_ITERS,
+                 profile: bool = False,
+                 writer=None):
         self.optimizer = optimizer
         self.group = group
         self.err_callback = err_callback
         self.warmup_iters = warmup_iters
         self.train_iter = 0
         self._synchronized = False
+        from ..utils.profiling import GpuPhaseTimer
+        self.phase_timer = GpuPhaseTimer(enabled=profile)
+        self.writer = writer  # SummaryWriter-compatible (add_scalar)
 
         if compression is None:
             compression = (TopKCompressor(density) if is_sparse
@@ -199,17 +204,30 @@ class _DistributedOptimizer(torch.optim.Optimizer):
             return
         world = self.reducer.world_size()
         warmup = self.train_iter < self.warmup_iters
+        pt = self.phase_timer
         if self.mode == "dense" or warmup:
+            pt.start("comm")
             self.reducer.dense(self.flat_grad)  # averaged in place
+            pt.stop("comm")
             self._staged = ("dense", None, None)
         elif self.mode == "gtopk":
+            pt.start("compress")
             vals, idx = self.compression.compress(self.flat_grad, "flat")
+            pt.stop("compress")
+            pt.start("comm")
             gval, gidx = self.reducer.gtopk(vals, idx)
+            pt.stop("comm")
+            pt.start("ef")
             self.compression.add_residuals(gidx, "flat")
+            pt.stop("ef")
             self._staged = ("sparse", gval, gidx)
         else:  # topk baseline: union sum sits in dense_scratch
+            pt.start("compress")
             vals, idx = self.compression.compress(self.flat_grad, "flat")
+            pt.stop("compress")
+            pt.start("comm")
             self.reducer.topk(vals, idx)
+            pt.stop("comm")
             # every local pick is included in the union → nothing rejected
             self._staged = ("union", None, None)
         self._synchronized = True
@@ -223,6 +241,7 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         kind, gval, gidx = self._staged
         world = self.reducer.world_size()
         g = self.optimizer.param_groups[0]
+        self.phase_timer.start("update")
         if self._fused:
             if kind == "dense":
                 grad_buf, scale, zero = self.flat_grad, 1.0, True
@@ -252,6 +271,10 @@ class _DistributedOptimizer(torch.optim.Optimizer):
             self.optimizer.step()
             if kind != "dense":
                 self.flat_grad.zero_()
+        self.phase_timer.stop("update")
+        if self.writer is not None and self.train_iter % 50 == 0:
+            for tag, ms in self.phase_timer.collect().items():
+                self.writer.add_scalar(f"opt/{tag}_ms", ms, self.train_iter)
         self.train_iter += 1
         self._synchronized = False
         return loss
@@ -282,10 +305,11 @@ def DistributedOptimizer(optimizer, named_parameters=None, compression=None,
                          is_sparse: bool = True, density: float = 0.001,
                          mode: Optional[str] = None, group=None,
                          err_callback=None,
-                         warmup_iters: int = WARMUP_ITERS):
+                         warmup_iters: int = WARMUP_ITERS,
+                         profile: bool = False, writer=None):
     """Factory mirroring the reference's
     ``DistributedOptimizer(optimizer, named_parameters, compression,
     is_sparse, density, err_callback, ...)``."""
     return _DistributedOptimizer(optimizer, named_parameters, compression,
                                  is_sparse, density, mode, group,
-                                 err_callback, warmup_iters)
+                                 err_callback, warmup_iters, profile, writer)

@@ -196,6 +196,19 @@ class DLTrainer:
         return lr
 
     # ------------------------------------------------------------------
+    @staticmethod
+    def _sync_metrics(metrics: dict) -> dict:
+        """C6: average scalar metrics across ranks when distributed."""
+        import torch.distributed as dist
+        if not (dist.is_available() and dist.is_initialized()) or \
+                dist.get_world_size() == 1:
+            return metrics
+        keys = sorted(metrics)
+        t = torch.tensor([float(metrics[k]) for k in keys])
+        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        t /= dist.get_world_size()
+        return {k: float(v) for k, v in zip(keys, t)}
+
     @torch.no_grad()
     def test(self, epoch: int = 0):
         """Top-1/top-5 for classification, perplexity for the LM,
@@ -217,7 +230,7 @@ class DLTrainer:
                 count += 1
             ppl = float(torch.exp(torch.tensor(total_loss / max(1, count))))
             logger.info("[epoch %d] test perplexity %.2f", epoch, ppl)
-            return {"perplexity": ppl}
+            return self._sync_metrics({"perplexity": ppl})
         if self.is_ctc:
             total_loss, count = 0.0, 0
             for batch in self.testloader:
@@ -226,7 +239,7 @@ class DLTrainer:
                 count += 1
             avg = total_loss / max(1, count)
             logger.info("[epoch %d] test ctc-loss %.4f", epoch, avg)
-            return {"ctc_loss": avg}
+            return self._sync_metrics({"ctc_loss": avg})
         top1, top5, total = 0, 0, 0
         for x, y in self.testloader:
             x = x.to(self.device)
@@ -239,7 +252,7 @@ class DLTrainer:
             total += y.numel()
         acc1, acc5 = top1 / max(1, total), top5 / max(1, total)
         logger.info("[epoch %d] test top1 %.4f top5 %.4f", epoch, acc1, acc5)
-        return {"top1": acc1, "top5": acc5}
+        return self._sync_metrics({"top1": acc1, "top5": acc5})
 
     # ------------------------------------------------------------------
     def save(self, prefix: str, run_tag: str, epoch: int, optimizer=None):

@@ -1,0 +1,41 @@
+"""train.py CLI end-to-end (single process, CPU, tiny)."""
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_train_cli_single_process(tmp_path):
+    metrics = str(tmp_path / "m.jsonl")
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "train.py"),
+         "--dnn", "mnistffn", "--dataset", "mnist", "--batch-size", "8",
+         "--max-epochs", "1", "--iters-per-epoch", "3", "--n-samples", "64",
+         "--density", "0.05", "--device", "cpu", "--no-amp",
+         "--num-workers", "0", "--metrics", metrics,
+         "--checkpoint-dir", str(tmp_path / "ckpt")],
+        cwd=REPO, capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-2000:]
+    rows = [json.loads(l) for l in open(metrics)]
+    assert rows and "samples_per_sec" in rows[0]
+    run_dirs = os.listdir(str(tmp_path / "ckpt"))
+    assert len(run_dirs) == 1
+    assert any(f.startswith("model-") for f in
+               os.listdir(os.path.join(str(tmp_path / "ckpt"), run_dirs[0])))
+
+
+def test_train_cli_gloo_world2(tmp_path):
+    env = dict(os.environ)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29517", os.path.join(REPO, "train.py"),
+         "--dnn", "mnistffn", "--dataset", "mnist", "--batch-size", "8",
+         "--max-epochs", "1", "--iters-per-epoch", "2", "--n-samples", "64",
+         "--density", "0.05", "--device", "cpu", "--no-amp",
+         "--num-workers", "0",
+         "--checkpoint-dir", str(tmp_path / "ckpt")],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, (r.stdout[-1000:], r.stderr[-2000:])

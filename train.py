@@ -111,7 +111,7 @@ def main():
                             "Speed %.1f samples/s/GPU", epoch, it + 1,
                             iters_per_epoch, loss, lr, seen / wall)
         wall = time.time() - t0
-        results = trainer.test(epoch) if rank == 0 else {}
+        results = trainer.test(epoch)  # all ranks: metrics are all-reduced
         if rank == 0:
             sps = iters_per_epoch * args.nsteps_update * args.batch_size / wall
             logger.info("epoch %d done in %.1fs — %.1f samples/s/GPU (%s)",
