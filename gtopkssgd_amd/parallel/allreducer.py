@@ -234,7 +234,9 @@ def _self_benchmark():  # pragma: no cover - exercised on GPU boxes
     k = max(1, int(args.density * n))
     comp = TopKCompressor(args.density)
     comp.residuals["b"] = torch.zeros(n, device=device)
-    reducer = GradReducer(n, k, device, mode=args.mode)
+    reducer = GradReducer(
+        n, k, device,
+        mode=args.mode if args.mode != "torchtopk" else "gtopk")
     g = torch.Generator(device="cpu").manual_seed(rank)
     grad = torch.randn(n, generator=g).to(device)
 
