@@ -61,7 +61,8 @@ class _DistributedOptimizer(torch.optim.Optimizer):
                  err_callback=None,
                  warmup_iters: int = WARMUP_ITERS,
                  profile: bool = False,
-                 writer=None):
+                 writer=None,
+                 overlap_dense: bool = True):
         self.optimizer = optimizer
         self.group = group
         self.err_callback = err_callback
@@ -143,6 +144,19 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         self.reducer = GradReducer(self.numel, self.k, device,
                                    mode=self.mode, group=group)
 
+        # --- dense-mode bucketed all-reduce overlapped with backward ---
+        # (used by dense mode and the dense warm-up iterations; the
+        # reference hid 1 Gbps Ethernet latency behind a comm thread —
+        # the xGMI-native equivalent is async RCCL launched from
+        # post-accumulate-grad hooks as 25 MB buckets complete.)
+        self.overlap_dense = overlap_dense
+        self._dense_handles = []
+        self._accum_boundary = True  # False during non-final micro-steps
+        self._buckets = []
+        if self.overlap_dense and \
+                (self.mode == "dense" or self.warmup_iters > 0):
+            self._setup_dense_overlap()
+
         self._broadcast_parameters()
         logger.info(
             "DistributedOptimizer: n=%d k=%d mode=%s fused=%s world=%d",
@@ -150,6 +164,79 @@ class _DistributedOptimizer(torch.optim.Optimizer):
 
     # `torch.optim.Optimizer.__init__` is intentionally NOT called; we
     # delegate param_groups/state to the inner optimizer.
+
+    # ------------------------------------------------------------------
+    # dense-overlap machinery
+    BUCKET_ELEMS = 25 * 1024 * 1024 // 4  # 25 MB fp32 buckets
+
+    def _setup_dense_overlap(self):
+        start, count, acc = 0, 0, 0
+        spans = []
+        for p, off in zip(self._params, self._offsets):
+            count += 1
+            acc += p.numel()
+            if acc >= self.BUCKET_ELEMS:
+                spans.append((start, off + p.numel(), count))
+                start, count, acc = off + p.numel(), 0, 0
+        if count:
+            spans.append((start, self.numel, count))
+        param_bucket = {}
+        pi = 0
+        for bi, (s, e, nparams) in enumerate(spans):
+            self._buckets.append(
+                {"start": s, "end": e, "nparams": nparams,
+                 "fired": 0, "issued": False})
+            for _ in range(nparams):
+                param_bucket[id(self._params[pi])] = bi
+                pi += 1
+        for p in self._params:
+            bi = param_bucket[id(p)]
+            p.register_post_accumulate_grad_hook(
+                lambda p_, bi=bi: self._on_grad_ready(bi))
+
+    def _overlap_active(self) -> bool:
+        return (self._accum_boundary and self._buckets and size() > 1 and
+                (self.mode == "dense" or
+                 self.train_iter < self.warmup_iters))
+
+    def _on_grad_ready(self, bi: int):
+        if not self._overlap_active():
+            return
+        b = self._buckets[bi]
+        b["fired"] += 1
+        if b["fired"] == b["nparams"]:
+            b["fired"] = 0
+            b["issued"] = True
+            view = self.flat_grad[b["start"]:b["end"]]
+            self._dense_handles.append(
+                dist.all_reduce(view, async_op=True, group=self.group))
+
+    def _finish_dense_overlap(self) -> bool:
+        """Wait outstanding bucket all-reduces; returns True when the
+        overlap path covered the whole gradient (else caller falls back
+        to the one-shot dense all-reduce)."""
+        if not self._dense_handles and \
+                not any(b["issued"] for b in self._buckets):
+            return False
+        for b in self._buckets:
+            if not b["issued"]:  # param never fired (unused this iter)
+                view = self.flat_grad[b["start"]:b["end"]]
+                self._dense_handles.append(
+                    dist.all_reduce(view, async_op=True, group=self.group))
+        for h in self._dense_handles:
+            h.wait()
+        self._dense_handles.clear()
+        for b in self._buckets:
+            b["issued"] = False
+            b["fired"] = 0
+        self.flat_grad.div_(self.reducer.world_size())
+        return True
+
+    def set_accumulation_boundary(self, is_boundary: bool):
+        """With nsteps_update > 1 gradient accumulation, call with False
+        before non-final micro-step backwards so dense-overlap hooks do
+        not all-reduce partial gradients; True before the final one."""
+        self._accum_boundary = bool(is_boundary)
 
     @staticmethod
     def _can_fuse(optimizer: torch.optim.Optimizer) -> bool:
@@ -207,7 +294,8 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         pt = self.phase_timer
         if self.mode == "dense" or warmup:
             pt.start("comm")
-            self.reducer.dense(self.flat_grad)  # averaged in place
+            if not self._finish_dense_overlap():
+                self.reducer.dense(self.flat_grad)  # averaged in place
             pt.stop("comm")
             self._staged = ("dense", None, None)
         elif self.mode == "gtopk":
@@ -306,10 +394,12 @@ def DistributedOptimizer(optimizer, named_parameters=None, compression=None,
                          mode: Optional[str] = None, group=None,
                          err_callback=None,
                          warmup_iters: int = WARMUP_ITERS,
-                         profile: bool = False, writer=None):
+                         profile: bool = False, writer=None,
+                         overlap_dense: bool = True):
     """Factory mirroring the reference's
     ``DistributedOptimizer(optimizer, named_parameters, compression,
     is_sparse, density, err_callback, ...)``."""
     return _DistributedOptimizer(optimizer, named_parameters, compression,
                                  is_sparse, density, mode, group,
-                                 err_callback, warmup_iters, profile, writer)
+                                 err_callback, warmup_iters, profile, writer,
+                                 overlap_dense)

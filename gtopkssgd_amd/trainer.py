@@ -153,7 +153,7 @@ class DLTrainer:
         gradients (grads land in the flat buffer via .grad views)."""
         self.net.train()
         total_loss = 0.0
-        for _ in range(num_of_iters):
+        for s in range(num_of_iters):
             self.timer.start("io")
             batch = self._next_batch()
             self.timer.stop("io")
@@ -163,6 +163,10 @@ class DLTrainer:
                 loss = self._forward_loss(batch)
             self.timer.stop("fwd")
             self.timer.start("bwd")
+            if dist_optimizer is not None and \
+                    hasattr(dist_optimizer, "set_accumulation_boundary"):
+                dist_optimizer.set_accumulation_boundary(
+                    s == num_of_iters - 1)
             (loss / self.nsteps_update).backward()
             self.timer.stop("bwd")
             total_loss += loss.item()

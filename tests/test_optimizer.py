@@ -178,3 +178,70 @@ def test_channels_last_params_preserved():
         memory_format=torch.channels_last)
     opt.step()
     assert not torch.equal(net[0].weight, w0)  # update applied
+
+
+def _worker_dense_overlap_parity(rank, world):
+    """Bucketed overlap path must produce the same result as the
+    one-shot dense all-reduce (rho irrelevant: mode=dense)."""
+    import gtopkssgd_amd.parallel.optimizer as om
+    om._DistributedOptimizer.BUCKET_ELEMS = 4096  # several buckets
+    results = {}
+    for overlap in (True, False):
+        model = _make_model()
+        inner = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
+        opt = gt.DistributedOptimizer(inner, model.named_parameters(),
+                                      compression=gt.compressors["none"](),
+                                      mode="dense", overlap_dense=overlap)
+        if overlap:
+            # tiny buckets so several fire per backward
+            assert opt._buckets
+        crit = nn.CrossEntropyLoss()
+        for s in range(3):
+            x, y = _batch(3000 + s + rank)
+            opt.zero_grad()
+            crit(model(x), y).backward()
+            opt.step()
+        results[overlap] = torch.cat([p.data.view(-1)
+                                      for p in model.parameters()])
+    assert torch.allclose(results[True], results[False], atol=1e-6), \
+        (results[True] - results[False]).abs().max().item()
+
+
+def test_dense_overlap_parity_world2():
+    run_in_procs(_worker_dense_overlap_parity, 2)
+
+
+def _worker_overlap_with_accumulation(rank, world):
+    """set_accumulation_boundary must keep partial gradients local."""
+    import gtopkssgd_amd.parallel.optimizer as om
+    om._DistributedOptimizer.BUCKET_ELEMS = 4096
+    model = _make_model()
+    inner = torch.optim.SGD(model.parameters(), lr=0.1)
+    opt = gt.DistributedOptimizer(inner, model.named_parameters(),
+                                  compression=gt.compressors["none"](),
+                                  mode="dense")
+    crit = nn.CrossEntropyLoss()
+    for s in range(2):
+        opt.zero_grad()
+        for micro in range(3):
+            opt.set_accumulation_boundary(micro == 2)
+            x, y = _batch(4000 + 10 * s + micro)
+            (crit(model(x), y) / 3).backward()
+        opt.step()
+    # reference: single process, grads averaged over micro-steps
+    ref = _make_model()
+    ropt = torch.optim.SGD(ref.parameters(), lr=0.1)
+    for s in range(2):
+        ropt.zero_grad()
+        for micro in range(3):
+            x, y = _batch(4000 + 10 * s + micro)
+            (crit(ref(x), y) / 3).backward()
+        ropt.step()
+    flat = torch.cat([p.data.view(-1) for p in model.parameters()])
+    rflat = torch.cat([p.data.view(-1) for p in ref.parameters()])
+    assert torch.allclose(flat, rflat, atol=1e-5), \
+        (flat - rflat).abs().max().item()
+
+
+def test_dense_overlap_accumulation_world2():
+    run_in_procs(_worker_overlap_with_accumulation, 2)
