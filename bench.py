@@ -68,8 +68,10 @@ def main():
     device = torch.device("cuda", local_rank)
     torch.manual_seed(1234)
     # MIOpen: exhaustive-find conv algo selection (amortised in warmup);
-    # without it MIOpen falls back to slow zero-workspace solvers.
-    torch.backends.cudnn.benchmark = True
+    # without it MIOpen's FAST fallback picks slow zero-workspace
+    # solvers.  GTOPK_CUDNN_BENCHMARK=0 disables for A/B testing.
+    torch.backends.cudnn.benchmark = \
+        os.environ.get("GTOPK_CUDNN_BENCHMARK", "1") == "1"
 
     model = resnet50().to(device).to(memory_format=torch.channels_last)
     inner = torch.optim.SGD(model.parameters(), lr=args.lr, momentum=0.9,
