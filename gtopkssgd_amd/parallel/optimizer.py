@@ -289,6 +289,14 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         staged for the update.  Idempotent per iteration."""
         if self._synchronized:
             return
+        try:
+            self._synchronize_impl()
+        except Exception as exc:  # reference API: err_callback on failure
+            if self.err_callback is not None:
+                self.err_callback(exc)
+            raise
+
+    def _synchronize_impl(self):
         world = self.reducer.world_size()
         warmup = self.train_iter < self.warmup_iters
         pt = self.phase_timer

@@ -55,6 +55,10 @@ def parse_args():
     p.add_argument("--checkpoint-dir", default="./checkpoints")
     p.add_argument("--run-tag", default=None)
     p.add_argument("--metrics", default=None, help="JSONL metrics path")
+    p.add_argument("--stats", default=None,
+                   help="JSONL per-phase optimizer stats path (rank 0)")
+    p.add_argument("--profile", action="store_true",
+                   help="enable HIP-event phase timers in the optimizer")
     return p.parse_args()
 
 
@@ -89,9 +93,14 @@ def main():
 
     compressor = gt.compressors[args.compression](args.density) \
         if args.compression != "none" else gt.compressors["none"]()
+    writer = None
+    if args.stats and rank == 0:
+        from gtopkssgd_amd.utils.profiling import StatsWriter
+        writer = StatsWriter(args.stats)
     opt = gt.DistributedOptimizer(
         trainer.optimizer, trainer.net.named_parameters(),
-        compression=compressor, density=args.density)
+        compression=compressor, density=args.density,
+        profile=args.profile or bool(writer), writer=writer)
 
     iters_per_epoch = args.iters_per_epoch or max(
         1, len(trainer.trainloader) // args.nsteps_update)
