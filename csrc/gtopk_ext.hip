@@ -39,6 +39,13 @@ __device__ __forceinline__ unsigned mag_key(float v) {
     return bits >> KEY_SHIFT;
 }
 
+// second-level key: the next 13 magnitude bits (bits 17..5) — the
+// two-level select is exact on the top 26 bits of |v|.
+__device__ __forceinline__ unsigned mag_key2(float v) {
+    unsigned bits = __float_as_uint(v) & 0x7fffffffu;
+    return (bits >> 5) & (HIST_BINS - 1);
+}
+
 // ---------------------------------------------------------------------------
 // K1+K2 pass 1: residual += grad, histogram of |residual| keys
 // ---------------------------------------------------------------------------
@@ -96,9 +103,12 @@ __global__ __launch_bounds__(BLOCK) void hist_vals_kernel(
 
 // ---------------------------------------------------------------------------
 // pass 2 (single block): find threshold bin T with
-//   count(bins > T) < k <= count(bins >= T); zero the compact counters.
-// info = [T, count_above]
+//   count(bins > T) < k <= count(bins >= T).
+// LEVEL=0: info = [T, count_above]; LEVEL=1 (within bin T): target is
+//   k - info[1]; writes info[2] = T2, info[3] = info[1] + count_above2
+//   and zeroes the compact counters.
 // ---------------------------------------------------------------------------
+template <int LEVEL>
 __global__ __launch_bounds__(BLOCK) void scan_threshold_kernel(
         const unsigned* __restrict__ hist, int k, int* __restrict__ info,
         int* __restrict__ counters) {
@@ -106,6 +116,7 @@ __global__ __launch_bounds__(BLOCK) void scan_threshold_kernel(
     const int t = threadIdx.x;
     const int chunk = HIST_BINS / BLOCK;        // 32 bins per thread
     const int base_hi = HIST_BINS - 1 - t * chunk;
+    const int target = (LEVEL == 0) ? k : (k - info[1]);
     unsigned sum = 0;
     for (int j = 0; j < chunk; ++j) sum += hist[base_hi - j];
     chunk_prefix[t] = sum;
@@ -117,18 +128,46 @@ __global__ __launch_bounds__(BLOCK) void scan_threshold_kernel(
             chunk_prefix[c] = acc;
             acc += s;
         }
-        counters[0] = 0;
-        counters[1] = 0;
+        if (LEVEL == 1) {
+            counters[0] = 0;
+            counters[1] = 0;
+        }
     }
     __syncthreads();
     unsigned cum = chunk_prefix[t];             // count strictly above chunk
     for (int j = 0; j < chunk; ++j) {
         const unsigned h = hist[base_hi - j];
-        if (cum < (unsigned)k && cum + h >= (unsigned)k) {
-            info[0] = base_hi - j;              // unique writer
-            info[1] = (int)cum;
+        if (cum < (unsigned)target && cum + h >= (unsigned)target) {
+            if (LEVEL == 0) {                   // unique writer
+                info[0] = base_hi - j;
+                info[1] = (int)cum;
+            } else {
+                info[2] = base_hi - j;
+                info[3] = info[1] + (int)cum;
+            }
         }
         cum += h;
+    }
+}
+
+// level-2 histogram over elements whose level-1 key == T
+__global__ __launch_bounds__(BLOCK) void refine_hist_kernel(
+        const float* __restrict__ src, long long n,
+        const int* __restrict__ info, unsigned* __restrict__ ghist2) {
+    __shared__ unsigned lhist[HIST_BINS];
+    for (int j = threadIdx.x; j < HIST_BINS; j += BLOCK) lhist[j] = 0;
+    __syncthreads();
+    const int T = info[0];
+    const long long stride = (long long)gridDim.x * BLOCK;
+    for (long long i = (long long)blockIdx.x * BLOCK + threadIdx.x; i < n;
+         i += stride) {
+        const float v = src[i];
+        if ((int)mag_key(v) == T) atomicAdd(&lhist[mag_key2(v)], 1u);
+    }
+    __syncthreads();
+    for (int j = threadIdx.x; j < HIST_BINS; j += BLOCK) {
+        unsigned c = lhist[j];
+        if (c) atomicAdd(&ghist2[j], c);
     }
 }
 
@@ -160,23 +199,27 @@ __global__ __launch_bounds__(BLOCK) void compact_kernel(
         long long n, const int* __restrict__ info, int* __restrict__ counters,
         float* __restrict__ out_val, int* __restrict__ out_idx, int k) {
     const int T = info[0];
-    const int above = info[1];
+    const int T2 = info[2];
+    const int above = info[3];   // count with (key,key2) strictly above (T,T2)
     const long long stride = (long long)gridDim.x * BLOCK;
     for (long long i = (long long)blockIdx.x * BLOCK + threadIdx.x; i < n;
          i += stride) {
         const float v = src[i];
         const int key = (int)mag_key(v);
+        const int key2 = (int)mag_key2(v);
+        const bool definite = key > T || (key == T && key2 > T2);
+        const bool tie = (key == T && key2 == T2);
         unsigned long long m;
-        // strict-above winners: guaranteed slots [0, above)
-        int slot = wave_alloc(&counters[0], key > T, &m);
-        if (key > T) {
+        // definite winners: guaranteed slots [0, above)
+        int slot = wave_alloc(&counters[0], definite, &m);
+        if (definite) {
             out_val[slot] = v;
             out_idx[slot] = HAS_IDX ? src_idx[i] : (int)i;
             if (ZERO_SRC) src[i] = 0.0f;
         }
-        // threshold-bin ties: slots [above, k)
-        slot = wave_alloc(&counters[1], key == T, &m);
-        if (key == T) {
+        // 26-bit ties: slots [above, k)
+        slot = wave_alloc(&counters[1], tie, &m);
+        if (tie) {
             slot += above;
             if (slot < k) {
                 out_val[slot] = v;
@@ -322,17 +365,21 @@ void ef_compress(at::Tensor residual, at::Tensor grad, int64_t k,
     CHECK_I32(hist); CHECK_I32(info); CHECK_I32(counters); CHECK_I32(out_idx);
     const long long n = residual.numel();
     TORCH_CHECK(grad.numel() == n && k > 0 && k <= n);
-    TORCH_CHECK(hist.numel() >= HIST_BINS && out_val.numel() >= k &&
+    TORCH_CHECK(hist.numel() >= 2 * HIST_BINS && out_val.numel() >= k &&
                 out_idx.numel() >= k);
     auto stream = cur_stream();
-    HIP_OK(hipMemsetAsync(hist.data_ptr<int>(), 0, HIST_BINS * sizeof(int), stream));
+    unsigned* h1 = (unsigned*)hist.data_ptr<int>();
+    unsigned* h2 = h1 + HIST_BINS;
+    HIP_OK(hipMemsetAsync(h1, 0, 2 * HIST_BINS * sizeof(int), stream));
     const int blocks = stream_blocks((n + 3) / 4);
     ef_accum_hist_kernel<<<blocks, BLOCK, 0, stream>>>(
-        residual.data_ptr<float>(), grad.data_ptr<float>(), n,
-        (unsigned*)hist.data_ptr<int>());
-    scan_threshold_kernel<<<1, BLOCK, 0, stream>>>(
-        (unsigned*)hist.data_ptr<int>(), (int)k, info.data_ptr<int>(),
-        counters.data_ptr<int>());
+        residual.data_ptr<float>(), grad.data_ptr<float>(), n, h1);
+    scan_threshold_kernel<0><<<1, BLOCK, 0, stream>>>(
+        h1, (int)k, info.data_ptr<int>(), counters.data_ptr<int>());
+    refine_hist_kernel<<<stream_blocks(n), BLOCK, 0, stream>>>(
+        residual.data_ptr<float>(), n, info.data_ptr<int>(), h2);
+    scan_threshold_kernel<1><<<1, BLOCK, 0, stream>>>(
+        h2, (int)k, info.data_ptr<int>(), counters.data_ptr<int>());
     compact_kernel<true, false><<<stream_blocks(n), BLOCK, 0, stream>>>(
         residual.data_ptr<float>(), nullptr, n, info.data_ptr<int>(),
         counters.data_ptr<int>(), out_val.data_ptr<float>(),
@@ -348,14 +395,20 @@ void topk_among(at::Tensor values, at::Tensor indexes, int64_t k,
     CHECK_I32(counters); CHECK_I32(out_idx);
     const long long m = values.numel();
     TORCH_CHECK(indexes.numel() == m && k > 0 && k <= m);
+    TORCH_CHECK(hist.numel() >= 2 * HIST_BINS);
     auto stream = cur_stream();
-    HIP_OK(hipMemsetAsync(hist.data_ptr<int>(), 0, HIST_BINS * sizeof(int), stream));
+    unsigned* h1 = (unsigned*)hist.data_ptr<int>();
+    unsigned* h2 = h1 + HIST_BINS;
+    HIP_OK(hipMemsetAsync(h1, 0, 2 * HIST_BINS * sizeof(int), stream));
     const int blocks = stream_blocks(m);
     hist_vals_kernel<<<blocks, BLOCK, 0, stream>>>(
-        values.data_ptr<float>(), m, (unsigned*)hist.data_ptr<int>());
-    scan_threshold_kernel<<<1, BLOCK, 0, stream>>>(
-        (unsigned*)hist.data_ptr<int>(), (int)k, info.data_ptr<int>(),
-        counters.data_ptr<int>());
+        values.data_ptr<float>(), m, h1);
+    scan_threshold_kernel<0><<<1, BLOCK, 0, stream>>>(
+        h1, (int)k, info.data_ptr<int>(), counters.data_ptr<int>());
+    refine_hist_kernel<<<blocks, BLOCK, 0, stream>>>(
+        values.data_ptr<float>(), m, info.data_ptr<int>(), h2);
+    scan_threshold_kernel<1><<<1, BLOCK, 0, stream>>>(
+        h2, (int)k, info.data_ptr<int>(), counters.data_ptr<int>());
     compact_kernel<false, true><<<blocks, BLOCK, 0, stream>>>(
         values.data_ptr<float>(), indexes.data_ptr<int>(), m,
         info.data_ptr<int>(), counters.data_ptr<int>(),

@@ -78,7 +78,8 @@ class _Scratch:
     """Per-device scratch tensors for the GPU select kernels."""
 
     def __init__(self, device: torch.device):
-        self.hist = torch.empty(HIST_BINS, dtype=torch.int32, device=device)
+        self.hist = torch.empty(2 * HIST_BINS, dtype=torch.int32,
+                                device=device)
         # info: [threshold_bin, count_above, k, pad]
         self.info = torch.empty(4, dtype=torch.int32, device=device)
         self.counters = torch.empty(2, dtype=torch.int32, device=device)

@@ -204,3 +204,26 @@ def test_gpu_density1_matches_dense_sgd():
                                   for p in model.parameters()]).cpu())
     assert torch.allclose(results[0], results[1], atol=1e-5), \
         (results[0] - results[1]).abs().max().item()
+
+
+def test_ef_compress_matches_torch_topk_random():
+    """Two-level (26-bit) select: on continuous random data the selected
+    index set must equal torch.topk's exactly (a 26-bit magnitude tie at
+    the k-boundary has ~0 probability)."""
+    torch.manual_seed(42)
+    for n, k in [(1_000_000, 1000), (250_000, 25_000), (4096, 4096)]:
+        g = (torch.randn(n) * torch.exp(torch.randn(n))).to(DEV)
+        res = torch.zeros(n, device=DEV)
+        v, i = ops.ef_compress(res, g, k)
+        ref = torch.topk(g.abs(), k).indices
+        assert set(i.tolist()) == set(ref.tolist()), (n, k)
+
+
+def test_topk_among_matches_torch_topk_random():
+    torch.manual_seed(43)
+    m, k = 51_114, 25_557  # 2k -> k at ResNet-50 rho=0.001 scale
+    val = torch.randn(m).to(DEV)
+    idx = torch.arange(m, dtype=torch.int32, device=DEV)
+    v, i = ops.topk_among(val, idx, k)
+    ref = torch.topk(val.abs(), k).indices
+    assert set(i.tolist()) == set(ref.tolist())
