@@ -285,3 +285,7 @@ def _self_benchmark():  # pragma: no cover - exercised on GPU boxes
 
 if __name__ == "__main__":  # pragma: no cover
     _self_benchmark()
+
+
+# Reference-familiar alias (`upstream:allreducer.py` class AllReducer).
+AllReducer = GradReducer
