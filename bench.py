@@ -58,13 +58,18 @@ def _prime_miopen_db():
 def main():
     args = parse_args()
     assert torch.cuda.is_available(), "bench.py requires an MI355X GPU"
-    _prime_miopen_db()
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     if world > 1:
         dist.init_process_group(backend="nccl")
         torch.cuda.set_device(local_rank)
+    # rank 0 extracts the MIOpen cache snapshot; others wait (a
+    # concurrent multi-rank extract would race on the same files)
+    if local_rank == 0:
+        _prime_miopen_db()
+    if world > 1:
+        dist.barrier()
     device = torch.device("cuda", local_rank)
     torch.manual_seed(1234)
     # MIOpen: exhaustive-find conv algo selection (amortised in warmup);
