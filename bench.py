@@ -3,6 +3,9 @@
 density=0.001, bf16 autocast — the BASELINE.json headline metric
 (images/sec whole job over N MI355X GPUs, weak scaling).
 
+Also runs the supporting BASELINE configs via --model
+(resnet20/vgg16 CIFAR-shape, lstm PTB-shape, alexnet).
+
 Single GPU:   python bench.py --gpus 1 --steps 30 --warmup 10
 Multi GPU:    python -m torch.distributed.run --nnodes=1 \
                  --nproc-per-node N --master-addr 127.0.0.1 bench.py \
@@ -20,7 +23,16 @@ import torch.distributed as dist  # noqa: E402
 import torch.nn as nn  # noqa: E402
 
 import gtopkssgd_amd as gt  # noqa: E402
-from gtopkssgd_amd.models import resnet50  # noqa: E402
+from gtopkssgd_amd import models as zoo  # noqa: E402
+
+# model -> (builder kwargs dataset-shape, default per-GPU batch, unit)
+_CONFIGS = {
+    "resnet50": dict(shape=(3, 224, 224), classes=1000, batch=256),
+    "alexnet": dict(shape=(3, 224, 224), classes=1000, batch=512),
+    "resnet20": dict(shape=(3, 32, 32), classes=10, batch=2048),
+    "vgg16": dict(shape=(3, 32, 32), classes=10, batch=1024),
+    "lstm": dict(seq_len=35, vocab=10000, batch=512),
+}
 
 
 def parse_args():
@@ -28,12 +40,14 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=10)
-    p.add_argument("--batch-size", type=int, default=256,
-                   help="per-GPU batch size (weak scaling)")
+    p.add_argument("--batch-size", type=int, default=0,
+                   help="per-GPU batch size (0 = model default)")
     p.add_argument("--density", type=float, default=0.001)
     p.add_argument("--mode", default="gtopk",
                    choices=["gtopk", "topk", "dense"])
-    p.add_argument("--model", default="resnet50")
+    p.add_argument("--model", default="resnet50",
+                   choices=sorted(_CONFIGS))
+    p.add_argument("--no-channels-last", action="store_true")
     p.add_argument("--lr", type=float, default=0.1)
     return p.parse_args()
 
@@ -78,7 +92,16 @@ def main():
     torch.backends.cudnn.benchmark = \
         os.environ.get("GTOPK_CUDNN_BENCHMARK", "1") == "1"
 
-    model = resnet50().to(device).to(memory_format=torch.channels_last)
+    cfg = _CONFIGS[args.model]
+    B = args.batch_size or cfg["batch"]
+    is_lm = args.model == "lstm"
+    channels_last = not is_lm and not args.no_channels_last
+    if is_lm:
+        model = zoo.lstm_ptb(cfg["vocab"], 1500).to(device)
+    else:
+        model = zoo.create_net(args.model).to(device)
+        if channels_last:
+            model = model.to(memory_format=torch.channels_last)
     inner = torch.optim.SGD(model.parameters(), lr=args.lr, momentum=0.9,
                             weight_decay=1e-4)
     comp = (gt.compressors["none"]() if args.mode == "dense"
@@ -87,19 +110,30 @@ def main():
                                   compression=comp, density=args.density)
     crit = nn.CrossEntropyLoss()
 
-    B = args.batch_size
     n_buf = 4  # pre-staged synthetic batches, cycled
     gen = torch.Generator(device="cpu").manual_seed(4321 + rank)
-    xs = [torch.randn(B, 3, 224, 224, generator=gen)
-          .to(device, memory_format=torch.channels_last) for _ in range(n_buf)]
-    ys = [torch.randint(0, 1000, (B,), generator=gen).to(device)
-          for _ in range(n_buf)]
+    if is_lm:
+        xs = [torch.randint(0, cfg["vocab"], (cfg["seq_len"] + 1, B),
+                            generator=gen).to(device) for _ in range(n_buf)]
+        hidden0 = model.init_hidden(B)
+    else:
+        xs = [torch.randn(B, *cfg["shape"], generator=gen)
+              .to(device, memory_format=(torch.channels_last if channels_last
+                                         else torch.contiguous_format))
+              for _ in range(n_buf)]
+        ys = [torch.randint(0, cfg["classes"], (B,), generator=gen).to(device)
+              for _ in range(n_buf)]
 
     def step(i):
-        x, y = xs[i % n_buf], ys[i % n_buf]
         opt.zero_grad()
         with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
-            loss = crit(model(x), y)
+            if is_lm:
+                toks = xs[i % n_buf]
+                logits, _ = model(toks[:-1], hidden0)
+                loss = crit(logits.reshape(-1, logits.size(-1)),
+                            toks[1:].reshape(-1))
+            else:
+                loss = crit(model(xs[i % n_buf]), ys[i % n_buf])
         loss.backward()
         opt.step()
         return loss
@@ -130,12 +164,19 @@ def main():
         elapsed = float(t.item())
 
     if rank == 0:
-        value = world * B * args.steps / elapsed
+        per_step_items = B * (cfg["seq_len"] if is_lm else 1)
+        value = world * per_step_items * args.steps / elapsed
+        unit = "tokens/sec" if is_lm else "images/sec"
+        metric = (f"{unit} (whole node) {args.model} {args.mode} "
+                  f"density={args.density}")
+        if args.model == "resnet50" and args.mode == "gtopk" \
+                and args.density == 0.001:
+            metric = ("images/sec (whole node) ResNet-50 gTop-k "
+                      "density=0.001")
         out = {
-            "metric": "images/sec (whole node) ResNet-50 gTop-k "
-                      "density=0.001",
+            "metric": metric,
             "value": round(value, 1),
-            "unit": "images/sec",
+            "unit": unit,
             "n_gpus": world,
             "steps": args.steps,
             "warmup": args.warmup,
