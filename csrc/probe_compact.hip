@@ -197,6 +197,84 @@ __global__ __launch_bounds__(BLOCK) void v4_blockagg(
     }
 }
 
+// v5: contiguous chunk per block; winners staged in LDS; ONE global
+// slot-claim atomic per flush (~1 per block) instead of one per wave.
+#define CAP 2048
+__global__ __launch_bounds__(BLOCK) void v5_ldsbuf(
+        float* src, long long n, const int* info, int* counters,
+        float* out_val, int* out_idx, int k) {
+    const int T = info[0], T2 = info[2], above = info[3];
+    __shared__ float lval[CAP];
+    __shared__ int lidx[CAP];
+    __shared__ float tval[CAP];
+    __shared__ int tidx[CAP];
+    __shared__ int lcnt, tcnt, gbase, tgbase;
+    if (threadIdx.x == 0) { lcnt = 0; tcnt = 0; }
+    __syncthreads();
+    const long long n4 = n >> 2;
+    const long long chunk = (n4 + gridDim.x - 1) / gridDim.x;
+    const long long c0 = (long long)blockIdx.x * chunk;
+    const long long c1 = c0 + chunk < n4 ? c0 + chunk : n4;
+    float4* s4 = reinterpret_cast<float4*>(src);
+    const long long iters = (chunk + BLOCK - 1) / BLOCK;
+    for (long long it = 0; it <= iters; ++it) {
+        if (it < iters) {
+            const long long i = c0 + it * BLOCK + threadIdx.x;
+            if (i < c1) {
+                float4 v4 = s4[i];
+                float vv[4] = {v4.x, v4.y, v4.z, v4.w};
+                bool z = false;
+                #pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    const float v = vv[j];
+                    const int key = (int)mag_key(v);
+                    const int key2 = (int)mag_key2(v);
+                    if (key > T || (key == T && key2 > T2)) {
+                        int p = atomicAdd(&lcnt, 1);
+                        lval[p] = v;
+                        lidx[p] = (int)(4 * i + j);
+                        vv[j] = 0.f;
+                        z = true;
+                    } else if (key == T && key2 == T2) {
+                        int p = atomicAdd(&tcnt, 1);
+                        tval[p] = v;
+                        tidx[p] = (int)(4 * i + j);
+                    }
+                }
+                if (z) s4[i] = make_float4(vv[0], vv[1], vv[2], vv[3]);
+            }
+        }
+        __syncthreads();
+        const bool last = (it == iters);
+        if ((lcnt > CAP - 4 * BLOCK) || (last && lcnt)) {
+            if (threadIdx.x == 0) gbase = atomicAdd(&counters[0], lcnt);
+            __syncthreads();
+            for (int p = threadIdx.x; p < lcnt; p += BLOCK) {
+                out_val[gbase + p] = lval[p];
+                out_idx[gbase + p] = lidx[p];
+            }
+            __syncthreads();
+            if (threadIdx.x == 0) lcnt = 0;
+        }
+        if ((tcnt > CAP - 4 * BLOCK) || (last && tcnt)) {
+            if (threadIdx.x == 0) tgbase = above + atomicAdd(&counters[1],
+                                                             tcnt);
+            __syncthreads();
+            for (int p = threadIdx.x; p < tcnt; p += BLOCK) {
+                const int slot = tgbase + p;
+                if (slot < k) {
+                    out_val[slot] = tval[p];
+                    out_idx[slot] = tidx[p];
+                    src[tidx[p]] = 0.f;
+                }
+            }
+            __syncthreads();
+            if (threadIdx.x == 0) tcnt = 0;
+        }
+        __syncthreads();
+    }
+}
+
 int main(int argc, char** argv) {
     long long n = argc > 1 ? atoll(argv[1]) : 25557032LL;
     int k = argc > 2 ? atoi(argv[2]) : 25557;
@@ -276,5 +354,28 @@ int main(int argc, char** argv) {
     bench("v4_blockagg", [&] {
         v4_blockagg<<<blocks4, BLOCK>>>(d_src, n, d_info, d_cnt,
                                         d_val, d_idx, k); });
+    bench("v5_ldsbuf", [&] {
+        v5_ldsbuf<<<blocks4, BLOCK>>>(d_src, n, d_info, d_cnt,
+                                      d_val, d_idx, k); });
+    // correctness spot-check for v5: def count exact, outputs unique
+    {
+        CHECK(hipMemcpy(d_src, h.data(), n * 4, hipMemcpyHostToDevice));
+        CHECK(hipMemset(d_cnt, 0, 8));
+        v5_ldsbuf<<<blocks4, BLOCK>>>(d_src, n, d_info, d_cnt, d_val,
+                                      d_idx, k);
+        std::vector<int> idx(k);
+        CHECK(hipMemcpy(idx.data(), d_idx, (size_t)k * 4,
+                        hipMemcpyDeviceToHost));
+        std::vector<char> seen(n, 0);
+        long long dup = 0, below = 0;
+        for (int key : idx) {
+            if (seen[key]) ++dup;
+            seen[key] = 1;
+            float av = fabsf(h[key]);
+            if (av < thr) ++below;
+        }
+        printf("v5 check: dup=%lld below_thr=%lld (expect 0, 0)\n",
+               dup, below);
+    }
     return 0;
 }
