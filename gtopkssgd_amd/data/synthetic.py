@@ -22,21 +22,35 @@ _SHAPES = {
 
 
 class SyntheticImages(Dataset):
-    """Random images + labels with a per-index deterministic generator."""
+    """Random images + labels with a per-index deterministic generator.
+
+    ``learnable=True`` draws each image around a fixed per-class mean
+    direction (SNR ~1), giving a task where loss/accuracy curves are
+    meaningful — used by the convergence-comparison harness
+    (benchmarks/convergence.py) to reproduce the reference's
+    gTop-k-vs-dense validation methodology without real datasets.
+    """
 
     def __init__(self, dataset: str = "cifar10", n_samples: int = 50000,
-                 seed: int = 0):
+                 seed: int = 0, learnable: bool = False):
         self.shape, self.num_classes = _SHAPES[dataset]
         self.n = n_samples
         self.seed = seed
+        self.learnable = learnable
+        if learnable:
+            g = torch.Generator().manual_seed(seed * 977 + 13)
+            self.class_means = torch.randn(
+                (self.num_classes,) + self.shape, generator=g)
 
     def __len__(self):
         return self.n
 
     def __getitem__(self, i) -> Tuple[torch.Tensor, int]:
         g = torch.Generator().manual_seed(self.seed * 1_000_003 + i)
-        x = torch.randn(self.shape, generator=g)
         y = int(torch.randint(0, self.num_classes, (1,), generator=g))
+        x = torch.randn(self.shape, generator=g)
+        if self.learnable:
+            x = x + self.class_means[y]
         return x, y
 
 
