@@ -69,6 +69,7 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         self.warmup_iters = warmup_iters
         self.train_iter = 0
         self._synchronized = False
+        self._grad_clean = False
         from ..utils.profiling import GpuPhaseTimer
         self.phase_timer = GpuPhaseTimer(enabled=profile)
         self.writer = writer  # SummaryWriter-compatible (add_scalar)
@@ -280,8 +281,21 @@ class _DistributedOptimizer(torch.optim.Optimizer):
                 dist.broadcast(p.data, src=0, group=self.group)
 
     def zero_grad(self, set_to_none: bool = False):
-        # views cannot be set to None — always zero the flat buffer
+        # views cannot be set to None — zero the flat buffer (skipped
+        # when the fused step already left it zeroed)
+        if self._grad_clean:
+            self._grad_clean = False
+            return
         self.flat_grad.zero_()
+
+    def clip_grad_norm_(self, max_norm: float) -> torch.Tensor:
+        """Whole-model grad clipping as ONE norm over the flat buffer
+        (equivalent to torch.nn.utils.clip_grad_norm_ over all params,
+        without per-tensor kernel launches)."""
+        total = torch.linalg.vector_norm(self.flat_grad)
+        scale = torch.clamp(max_norm / (total + 1e-6), max=1.0)
+        self.flat_grad.mul_(scale)  # no host sync: clamp keeps <=1
+        return total
 
     # ------------------------------------------------------------------
     def synchronize(self):
@@ -338,6 +352,7 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         world = self.reducer.world_size()
         g = self.optimizer.param_groups[0]
         self.phase_timer.start("update")
+        self._grad_clean = False
         if self._fused:
             if kind == "dense":
                 grad_buf, scale, zero = self.flat_grad, 1.0, True
@@ -355,6 +370,9 @@ class _DistributedOptimizer(torch.optim.Optimizer):
                           weight_decay=g.get("weight_decay", 0.0),
                           dampening=0.0, nesterov=g.get("nesterov", False),
                           zero_grad=zero, grad_scale=scale)
+            # flat_grad ends zeroed on every fused path (dense: the
+            # kernel's self-zeroing; sparse/union: explicit zero above)
+            self._grad_clean = True
         else:
             if kind == "sparse":
                 self.flat_grad.zero_()
@@ -367,6 +385,7 @@ class _DistributedOptimizer(torch.optim.Optimizer):
             self.optimizer.step()
             if kind != "dense":
                 self.flat_grad.zero_()
+                self._grad_clean = True
         self.phase_timer.stop("update")
         if self.writer is not None and self.train_iter % 50 == 0:
             for tag, ms in self.phase_timer.collect().items():

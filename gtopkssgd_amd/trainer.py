@@ -180,8 +180,11 @@ class DLTrainer:
         opt = dist_optimizer if dist_optimizer is not None else self.optimizer
         self.timer.start("update")
         if self.norm_clip is not None:
-            torch.nn.utils.clip_grad_norm_(self.net.parameters(),
-                                           self.norm_clip)
+            if hasattr(opt, "clip_grad_norm_"):
+                opt.clip_grad_norm_(self.norm_clip)  # one flat-norm kernel
+            else:
+                torch.nn.utils.clip_grad_norm_(self.net.parameters(),
+                                               self.norm_clip)
         opt.step()
         opt.zero_grad()
         self.timer.stop("update")
@@ -263,6 +266,14 @@ class DLTrainer:
         if self.rank == 0:
             return save_checkpoint(prefix, run_tag, epoch, self.net,
                                    optimizer)
+
+    def get_acc(self):
+        """Reference parity: running average training loss since the
+        last reset (`upstream:dl_trainer.py` get_acc)."""
+        avg = self.avg_loss / max(1, self._loss_count)
+        self.avg_loss = 0.0
+        self._loss_count = 0
+        return avg
 
     def speed_summary(self, iters: int, wall: float) -> str:
         sps = iters * self.batch_size / wall
