@@ -541,7 +541,25 @@ void sgd_step(at::Tensor param, at::Tensor mom, at::Tensor grad, double lr,
     check_launch("gtopk");
 }
 
+// fused BatchNorm (csrc/fusedbn.hip)
+std::vector<at::Tensor> bn_fwd_train(
+    at::Tensor x, c10::optional<at::Tensor> z, at::Tensor gamma,
+    at::Tensor beta, at::Tensor running_mean, at::Tensor running_var,
+    double momentum, double eps, bool relu, at::Tensor stats_scratch);
+at::Tensor bn_fwd_eval(at::Tensor x, c10::optional<at::Tensor> z,
+                       at::Tensor gamma, at::Tensor beta,
+                       at::Tensor running_mean, at::Tensor running_var,
+                       double eps, bool relu);
+std::vector<at::Tensor> bn_bwd(
+    at::Tensor dy, at::Tensor x, c10::optional<at::Tensor> y,
+    at::Tensor gamma, at::Tensor saved_mean, at::Tensor saved_rstd,
+    bool relu, bool need_dz, at::Tensor stats_scratch);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("bn_fwd_train", &bn_fwd_train,
+          "fused NHWC BN(+add)(+ReLU) training forward (gfx950)");
+    m.def("bn_fwd_eval", &bn_fwd_eval, "fused NHWC BN eval forward");
+    m.def("bn_bwd", &bn_bwd, "fused NHWC BN(+add)(+ReLU) backward");
     m.def("ef_compress", &ef_compress,
           "fused residual accumulate + bucketed top-k select (gfx950)");
     m.def("topk_among", &topk_among, "top-k among candidate (val,idx) pairs");

@@ -9,6 +9,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.fusedbn import FusedBNAct
+
 
 class LambdaPad(nn.Module):
     """Option-A shortcut: stride-2 subsample + zero-pad channels."""
@@ -20,7 +22,10 @@ class LambdaPad(nn.Module):
     def forward(self, x):
         out = x[:, :, ::2, ::2]
         pad = self.planes_out - out.size(1)
-        return F.pad(out, (0, 0, 0, 0, pad // 2, pad - pad // 2))
+        out = F.pad(out, (0, 0, 0, 0, pad // 2, pad - pad // 2))
+        if out.is_cuda:  # keep the fused-BN residual path applicable
+            out = out.contiguous(memory_format=torch.channels_last)
+        return out
 
 
 class BasicBlock(nn.Module):
@@ -30,19 +35,18 @@ class BasicBlock(nn.Module):
         super().__init__()
         self.conv1 = nn.Conv2d(in_planes, planes, 3, stride=stride,
                                padding=1, bias=False)
-        self.bn1 = nn.BatchNorm2d(planes)
+        self.bn1 = FusedBNAct(planes, relu=True)
         self.conv2 = nn.Conv2d(planes, planes, 3, stride=1, padding=1,
                                bias=False)
-        self.bn2 = nn.BatchNorm2d(planes)
+        self.bn2 = FusedBNAct(planes, relu=True)  # relu after the add
         if stride != 1 or in_planes != planes:
             self.shortcut = LambdaPad(planes)
         else:
             self.shortcut = nn.Identity()
 
     def forward(self, x):
-        out = F.relu(self.bn1(self.conv1(x)))
-        out = self.bn2(self.conv2(out))
-        return F.relu(out + self.shortcut(x))
+        out = self.bn1(self.conv1(x))
+        return self.bn2(self.conv2(out), residual=self.shortcut(x))
 
 
 class CifarResNet(nn.Module):
@@ -50,7 +54,7 @@ class CifarResNet(nn.Module):
         super().__init__()
         self.in_planes = 16
         self.conv1 = nn.Conv2d(3, 16, 3, stride=1, padding=1, bias=False)
-        self.bn1 = nn.BatchNorm2d(16)
+        self.bn1 = FusedBNAct(16, relu=True)
         self.layer1 = self._make_layer(16, num_blocks, 1)
         self.layer2 = self._make_layer(32, num_blocks, 2)
         self.layer3 = self._make_layer(64, num_blocks, 2)
@@ -64,7 +68,7 @@ class CifarResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x):
-        out = F.relu(self.bn1(self.conv1(x)))
+        out = self.bn1(self.conv1(x))
         out = self.layer3(self.layer2(self.layer1(out)))
         out = F.adaptive_avg_pool2d(out, 1).flatten(1)
         return self.linear(out)

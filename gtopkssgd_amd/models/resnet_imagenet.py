@@ -9,6 +9,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.fusedbn import FusedBNAct
+
 
 class Bottleneck(nn.Module):
     expansion = 4
@@ -16,23 +18,23 @@ class Bottleneck(nn.Module):
     def __init__(self, in_planes, planes, stride=1, downsample=None):
         super().__init__()
         self.conv1 = nn.Conv2d(in_planes, planes, 1, bias=False)
-        self.bn1 = nn.BatchNorm2d(planes)
+        self.bn1 = FusedBNAct(planes, relu=True)
         self.conv2 = nn.Conv2d(planes, planes, 3, stride=stride, padding=1,
                                bias=False)
-        self.bn2 = nn.BatchNorm2d(planes)
+        self.bn2 = FusedBNAct(planes, relu=True)
         self.conv3 = nn.Conv2d(planes, planes * 4, 1, bias=False)
-        self.bn3 = nn.BatchNorm2d(planes * 4)
+        self.bn3 = FusedBNAct(planes * 4, relu=True)  # relu after add
         self.downsample = downsample
         self.stride = stride
 
     def forward(self, x):
         identity = x
-        out = F.relu(self.bn1(self.conv1(x)), inplace=True)
-        out = F.relu(self.bn2(self.conv2(out)), inplace=True)
-        out = self.bn3(self.conv3(out))
+        out = self.bn1(self.conv1(x))
+        out = self.bn2(self.conv2(out))
+        out = self.conv3(out)
         if self.downsample is not None:
             identity = self.downsample(x)
-        return F.relu(out + identity, inplace=True)
+        return self.bn3(out, residual=identity)
 
 
 class ResNet(nn.Module):
@@ -40,7 +42,7 @@ class ResNet(nn.Module):
         super().__init__()
         self.in_planes = 64
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = nn.BatchNorm2d(64)
+        self.bn1 = FusedBNAct(64, relu=True)
         self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         self.layer1 = self._make_layer(64, layers[0])
         self.layer2 = self._make_layer(128, layers[1], stride=2)
@@ -61,7 +63,7 @@ class ResNet(nn.Module):
             downsample = nn.Sequential(
                 nn.Conv2d(self.in_planes, planes * 4, 1, stride=stride,
                           bias=False),
-                nn.BatchNorm2d(planes * 4))
+                FusedBNAct(planes * 4))
         layers = [Bottleneck(self.in_planes, planes, stride, downsample)]
         self.in_planes = planes * 4
         for _ in range(1, blocks):
@@ -69,7 +71,7 @@ class ResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x):
-        x = self.maxpool(F.relu(self.bn1(self.conv1(x)), inplace=True))
+        x = self.maxpool(self.bn1(self.conv1(x)))
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         x = F.adaptive_avg_pool2d(x, 1).flatten(1)
         return self.fc(x)

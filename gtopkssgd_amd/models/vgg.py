@@ -5,6 +5,8 @@ the standard cfg dict; written from the VGG paper's layer spec.
 """
 import torch.nn as nn
 
+from ..ops.fusedbn import FusedBNAct
+
 cfg = {
     "VGG11": [64, "M", 128, "M", 256, 256, "M", 512, 512, "M", 512, 512, "M"],
     "VGG13": [64, 64, "M", 128, 128, "M", 256, 256, "M", 512, 512, "M",
@@ -30,7 +32,7 @@ class VGG(nn.Module):
                 layers.append(nn.MaxPool2d(2, 2))
             else:
                 layers += [nn.Conv2d(in_ch, x, 3, padding=1, bias=False),
-                           nn.BatchNorm2d(x), nn.ReLU(inplace=True)]
+                           FusedBNAct(x, relu=True)]
                 in_ch = x
         layers.append(nn.AvgPool2d(1, 1))
         return nn.Sequential(*layers)

@@ -23,7 +23,7 @@ setup(
     ext_modules=[
         CUDAExtension(
             name="gtopkssgd_amd._C",
-            sources=["csrc/gtopk_ext.hip"],
+            sources=["csrc/gtopk_ext.hip", "csrc/fusedbn.hip"],
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "--offload-arch=gfx950"],
