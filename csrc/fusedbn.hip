@@ -102,9 +102,19 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_stats_kernel(
             }
         }
     }
-    // block tree-reduction over threads sharing a chunk (stride ch_chunks)
-    __shared__ float red[BN_BLOCK];
+    // block tree-reduction over threads sharing a chunk (stride
+    // ch_chunks).  When ch_chunks >= BN_BLOCK each thread's chunk is
+    // unique within the block — skip the tree and emit directly.
     const int t = threadIdx.x;
+    if (ch_chunks >= BN_BLOCK) {
+        #pragma unroll
+        for (int i = 0; i < W; ++i) {
+            if (s[i] != 0.0f) atomicAdd(&sums[chunk * W + i], s[i]);
+            if (q[i] != 0.0f) atomicAdd(&sums[c + chunk * W + i], q[i]);
+        }
+        return;
+    }
+    __shared__ float red[BN_BLOCK];
     #pragma unroll
     for (int i = 0; i < W; ++i) {
         red[t] = s[i];
@@ -223,8 +233,16 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_bwd_stats_kernel(
             }
         }
     }
-    __shared__ float red[BN_BLOCK];
     const int t = threadIdx.x;
+    if (ch_chunks >= BN_BLOCK) {
+        #pragma unroll
+        for (int i = 0; i < W; ++i) {
+            if (db[i] != 0.0f) atomicAdd(&sums[chunk * W + i], db[i]);
+            if (dg[i] != 0.0f) atomicAdd(&sums[c + chunk * W + i], dg[i]);
+        }
+        return;
+    }
+    __shared__ float red[BN_BLOCK];
     #pragma unroll
     for (int i = 0; i < W; ++i) {
         red[t] = db[i];
