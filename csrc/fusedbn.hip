@@ -79,21 +79,35 @@ template <> struct VecTraits<float> {
 };
 
 // ---- stats: per-channel sum / sumsq ---------------------------------------
+// Row mapping (all data kernels): each block owns a CONTIGUOUS row
+// range [r_begin, r_end) — a global round-robin row stride makes every
+// wave access open a fresh DRAM page (measured 0.25 TB/s); contiguous
+// block ranges stream at full bandwidth.  Within a block, thread t owns
+// channel-chunk t % ch_chunks and walks rows at stride BLOCK/ch_chunks
+// (requires ch_chunks <= BN_BLOCK; the fp32 C=2048 case falls back to
+// a global-stride loop).
+#define BN_ROW_RANGE(first_stmt) \
+    const long long rows_per_block = (m + gridDim.x - 1) / gridDim.x; \
+    const long long r_begin = (long long)blockIdx.x * rows_per_block; \
+    const long long r_end = \
+        (r_begin + rows_per_block < m) ? r_begin + rows_per_block : m; \
+    first_stmt
+
 template <typename T>
 __global__ __launch_bounds__(BN_BLOCK) void bn_stats_kernel(
         const T* __restrict__ x, long long m, int c,
-        float* __restrict__ sums /* [2C]: sum | sumsq */) {
+        float* __restrict__ partials /* [gridDim][2C], zero-init */) {
     constexpr int W = VecTraits<T>::W;
     const int ch_chunks = c / W;
-    const int tid = blockIdx.x * BN_BLOCK + threadIdx.x;
-    const int chunk = tid % ch_chunks;          // fixed channel chunk
-    const long long r0 = tid / ch_chunks;
-    const long long rstride =
-        ((long long)gridDim.x * BN_BLOCK) / ch_chunks;
     float s[W] = {0}, q[W] = {0};
     float v[W];
-    if (rstride > 0) {
-        for (long long r = r0; r < m; r += rstride) {
+    int chunk;
+    if (ch_chunks <= BN_BLOCK) {
+        chunk = threadIdx.x % ch_chunks;
+        BN_ROW_RANGE(;)
+        const int rstep = BN_BLOCK / ch_chunks;
+        for (long long r = r_begin + threadIdx.x / ch_chunks; r < r_end;
+             r += rstep) {
             VecTraits<T>::load(x + r * c + chunk * W, v);
             #pragma unroll
             for (int i = 0; i < W; ++i) {
@@ -101,16 +115,32 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_stats_kernel(
                 q[i] += v[i] * v[i];
             }
         }
+    } else {  // rare fp32 large-C fallback
+        const int tid = blockIdx.x * BN_BLOCK + threadIdx.x;
+        chunk = tid % ch_chunks;
+        const long long rstride =
+            ((long long)gridDim.x * BN_BLOCK) / ch_chunks;
+        if (rstride > 0) {
+            for (long long r = tid / ch_chunks; r < m; r += rstride) {
+                VecTraits<T>::load(x + r * c + chunk * W, v);
+                #pragma unroll
+                for (int i = 0; i < W; ++i) {
+                    s[i] += v[i];
+                    q[i] += v[i] * v[i];
+                }
+            }
+        }
     }
-    // block tree-reduction over threads sharing a chunk (stride
-    // ch_chunks).  When ch_chunks >= BN_BLOCK each thread's chunk is
-    // unique within the block — skip the tree and emit directly.
+    // block tree-reduction over threads sharing a chunk, then ONE
+    // non-atomic partial-row store per block (a per-channel atomicAdd
+    // builds gridDim-long same-address chains — measured 100s of µs).
+    float* mine = partials + (long long)blockIdx.x * 2 * c;
     const int t = threadIdx.x;
-    if (ch_chunks >= BN_BLOCK) {
+    if (ch_chunks >= BN_BLOCK) {  // unique chunk per thread in block
         #pragma unroll
         for (int i = 0; i < W; ++i) {
-            if (s[i] != 0.0f) atomicAdd(&sums[chunk * W + i], s[i]);
-            if (q[i] != 0.0f) atomicAdd(&sums[c + chunk * W + i], q[i]);
+            mine[chunk * W + i] = s[i];
+            mine[c + chunk * W + i] = q[i];
         }
         return;
     }
@@ -123,8 +153,7 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_stats_kernel(
             if (t < off) red[t] += red[t + off];
             __syncthreads();
         }
-        if (t < ch_chunks && red[t] != 0.0f)
-            atomicAdd(&sums[t * W + i], red[t]);
+        if (t < ch_chunks) mine[t * W + i] = red[t];
         __syncthreads();
         red[t] = q[i];
         __syncthreads();
@@ -132,9 +161,21 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_stats_kernel(
             if (t < off) red[t] += red[t + off];
             __syncthreads();
         }
-        if (t < ch_chunks && red[t] != 0.0f)
-            atomicAdd(&sums[c + t * W + i], red[t]);
+        if (t < ch_chunks) mine[c + t * W + i] = red[t];
         __syncthreads();
+    }
+}
+
+// ---- reduce per-block partials into sums[2C] ------------------------------
+__global__ __launch_bounds__(BN_BLOCK) void bn_reduce_partials_kernel(
+        const float* __restrict__ partials, int nblocks, int c,
+        float* __restrict__ sums) {
+    for (int i = blockIdx.x * BN_BLOCK + threadIdx.x; i < 2 * c;
+         i += gridDim.x * BN_BLOCK) {
+        float acc = 0.0f;
+        for (int b = 0; b < nblocks; ++b)
+            acc += partials[(long long)b * 2 * c + i];
+        sums[i] = acc;
     }
 }
 
@@ -169,12 +210,22 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_fwd_kernel(
         const float* __restrict__ gamma, const float* __restrict__ beta) {
     constexpr int W = VecTraits<T>::W;
     const int ch_chunks = c / W;
-    const int tid = blockIdx.x * BN_BLOCK + threadIdx.x;
-    const int chunk = tid % ch_chunks;
-    const long long r0 = tid / ch_chunks;
-    const long long rstride =
-        ((long long)gridDim.x * BN_BLOCK) / ch_chunks;
-    if (rstride == 0) return;
+    long long r0, r_limit, rstep;
+    int chunk;
+    if (ch_chunks <= BN_BLOCK) {
+        BN_ROW_RANGE(;)
+        chunk = threadIdx.x % ch_chunks;
+        r0 = r_begin + threadIdx.x / ch_chunks;
+        r_limit = r_end;
+        rstep = BN_BLOCK / ch_chunks;
+    } else {
+        const int tid = blockIdx.x * BN_BLOCK + threadIdx.x;
+        chunk = tid % ch_chunks;
+        r0 = tid / ch_chunks;
+        r_limit = m;
+        rstep = ((long long)gridDim.x * BN_BLOCK) / ch_chunks;
+        if (rstep == 0) return;
+    }
     float sc[W], sh[W];
     #pragma unroll
     for (int i = 0; i < W; ++i) {
@@ -183,7 +234,7 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_fwd_kernel(
         sh[i] = beta[ci] - mean[ci] * sc[i];
     }
     float v[W], zr[W];
-    for (long long r = r0; r < m; r += rstride) {
+    for (long long r = r0; r < r_limit; r += rstep) {
         VecTraits<T>::load(x + r * c + chunk * W, v);
         if (ADD) VecTraits<T>::load(z + r * c + chunk * W, zr);
         #pragma unroll
@@ -203,14 +254,24 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_bwd_stats_kernel(
         const T* __restrict__ dy, const T* __restrict__ x,
         const T* __restrict__ y, long long m, int c,
         const float* __restrict__ mean, const float* __restrict__ rstd,
-        float* __restrict__ sums /* [2C]: dbeta | dgamma */) {
+        float* __restrict__ partials /* [gridDim][2C], zero-init */) {
     constexpr int W = VecTraits<T>::W;
     const int ch_chunks = c / W;
-    const int tid = blockIdx.x * BN_BLOCK + threadIdx.x;
-    const int chunk = tid % ch_chunks;
-    const long long r0 = tid / ch_chunks;
-    const long long rstride =
-        ((long long)gridDim.x * BN_BLOCK) / ch_chunks;
+    long long r0, r_limit, rstep;
+    int chunk;
+    if (ch_chunks <= BN_BLOCK) {
+        BN_ROW_RANGE(;)
+        chunk = threadIdx.x % ch_chunks;
+        r0 = r_begin + threadIdx.x / ch_chunks;
+        r_limit = r_end;
+        rstep = BN_BLOCK / ch_chunks;
+    } else {
+        const int tid = blockIdx.x * BN_BLOCK + threadIdx.x;
+        chunk = tid % ch_chunks;
+        r0 = tid / ch_chunks;
+        r_limit = m;
+        rstep = ((long long)gridDim.x * BN_BLOCK) / ch_chunks;
+    }
     float mu[W], rs[W];
     #pragma unroll
     for (int i = 0; i < W; ++i) {
@@ -219,8 +280,8 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_bwd_stats_kernel(
     }
     float db[W] = {0}, dg[W] = {0};
     float g[W], xv[W], yv[W];
-    if (rstride > 0) {
-        for (long long r = r0; r < m; r += rstride) {
+    if (rstep > 0) {
+        for (long long r = r0; r < r_limit; r += rstep) {
             const long long base = r * c + chunk * W;
             VecTraits<T>::load(dy + base, g);
             VecTraits<T>::load(x + base, xv);
@@ -233,12 +294,13 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_bwd_stats_kernel(
             }
         }
     }
+    float* mine = partials + (long long)blockIdx.x * 2 * c;
     const int t = threadIdx.x;
     if (ch_chunks >= BN_BLOCK) {
         #pragma unroll
         for (int i = 0; i < W; ++i) {
-            if (db[i] != 0.0f) atomicAdd(&sums[chunk * W + i], db[i]);
-            if (dg[i] != 0.0f) atomicAdd(&sums[c + chunk * W + i], dg[i]);
+            mine[chunk * W + i] = db[i];
+            mine[c + chunk * W + i] = dg[i];
         }
         return;
     }
@@ -251,8 +313,7 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_bwd_stats_kernel(
             if (t < off) red[t] += red[t + off];
             __syncthreads();
         }
-        if (t < ch_chunks && red[t] != 0.0f)
-            atomicAdd(&sums[t * W + i], red[t]);
+        if (t < ch_chunks) mine[t * W + i] = red[t];
         __syncthreads();
         red[t] = dg[i];
         __syncthreads();
@@ -260,8 +321,7 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_bwd_stats_kernel(
             if (t < off) red[t] += red[t + off];
             __syncthreads();
         }
-        if (t < ch_chunks && red[t] != 0.0f)
-            atomicAdd(&sums[c + t * W + i], red[t]);
+        if (t < ch_chunks) mine[c + t * W + i] = red[t];
         __syncthreads();
     }
 }
@@ -276,12 +336,22 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_bwd_dx_kernel(
         const float* __restrict__ sums /* dbeta | dgamma */) {
     constexpr int W = VecTraits<T>::W;
     const int ch_chunks = c / W;
-    const int tid = blockIdx.x * BN_BLOCK + threadIdx.x;
-    const int chunk = tid % ch_chunks;
-    const long long r0 = tid / ch_chunks;
-    const long long rstride =
-        ((long long)gridDim.x * BN_BLOCK) / ch_chunks;
-    if (rstride == 0) return;
+    long long r0, r_limit, rstep;
+    int chunk;
+    if (ch_chunks <= BN_BLOCK) {
+        BN_ROW_RANGE(;)
+        chunk = threadIdx.x % ch_chunks;
+        r0 = r_begin + threadIdx.x / ch_chunks;
+        r_limit = r_end;
+        rstep = BN_BLOCK / ch_chunks;
+    } else {
+        const int tid = blockIdx.x * BN_BLOCK + threadIdx.x;
+        chunk = tid % ch_chunks;
+        r0 = tid / ch_chunks;
+        r_limit = m;
+        rstep = ((long long)gridDim.x * BN_BLOCK) / ch_chunks;
+        if (rstep == 0) return;
+    }
     const float inv_m = 1.0f / (float)m;
     float mu[W], rs[W], gm[W], dbm[W], dgm[W];
     #pragma unroll
@@ -294,7 +364,7 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_bwd_dx_kernel(
         dgm[i] = sums[c + ci] * inv_m;
     }
     float g[W], xv[W], yv[W], dzv[W];
-    for (long long r = r0; r < m; r += rstride) {
+    for (long long r = r0; r < r_limit; r += rstep) {
         const long long base = r * c + chunk * W;
         VecTraits<T>::load(dy + base, g);
         VecTraits<T>::load(x + base, xv);
@@ -383,17 +453,20 @@ std::vector<at::Tensor> bn_fwd_train(
     auto saved_mean = at::empty({c}, gamma.options());
     auto saved_rstd = at::empty({c}, gamma.options());
     float* sums = stats_scratch.data_ptr<float>();
-    BN_OK(hipMemsetAsync(sums, 0, 2 * c * sizeof(float), stream));
     const int ch = c / W;
+    const int grid = bn_grid(m, ch);
+    auto partials = at::zeros({grid, 2 * c}, gamma.options());
     if (x.scalar_type() == at::kBFloat16) {
-        bn_stats_kernel<__hip_bfloat16>
-            <<<bn_grid(m, ch), BN_BLOCK, 0, stream>>>(
-                reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()), m, c,
-                sums);
+        bn_stats_kernel<__hip_bfloat16><<<grid, BN_BLOCK, 0, stream>>>(
+            reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()), m, c,
+            partials.data_ptr<float>());
     } else {
-        bn_stats_kernel<float><<<bn_grid(m, ch), BN_BLOCK, 0, stream>>>(
-            x.data_ptr<float>(), m, c, sums);
+        bn_stats_kernel<float><<<grid, BN_BLOCK, 0, stream>>>(
+            x.data_ptr<float>(), m, c, partials.data_ptr<float>());
     }
+    bn_reduce_partials_kernel<<<(2 * c + BN_BLOCK - 1) / BN_BLOCK,
+                                BN_BLOCK, 0, stream>>>(
+        partials.data_ptr<float>(), grid, c, sums);
     bn_finalize_kernel<<<(c + BN_BLOCK - 1) / BN_BLOCK, BN_BLOCK, 0,
                          stream>>>(
         sums, m, c, (float)eps, (float)momentum,
@@ -444,11 +517,12 @@ std::vector<at::Tensor> bn_bwd(
     auto dx = at::empty_like(x);
     auto dz = need_dz ? at::empty_like(x) : at::Tensor();
     float* sums = stats_scratch.data_ptr<float>();
-    BN_OK(hipMemsetAsync(sums, 0, 2 * c * sizeof(float), stream));
     const int W = x.scalar_type() == at::kBFloat16 ? 8 : 4;
     const int ch = c / W;
     TORCH_CHECK(c % W == 0 && (ch & (ch - 1)) == 0, "bad C");
     const int grid = bn_grid(m, ch);
+    auto partials = at::zeros({grid, 2 * c}, gamma.options());
+    float* pp = partials.data_ptr<float>();
     TORCH_CHECK(!relu || y.has_value(), "relu backward needs y");
 
     #define DISPATCH_BWD(T, TPTR) do { \
@@ -461,11 +535,14 @@ std::vector<at::Tensor> bn_bwd(
         if (relu) \
             bn_bwd_stats_kernel<T, true><<<grid, BN_BLOCK, 0, stream>>>( \
                 dyp, xp, yp, m, c, saved_mean.data_ptr<float>(), \
-                saved_rstd.data_ptr<float>(), sums); \
+                saved_rstd.data_ptr<float>(), pp); \
         else \
             bn_bwd_stats_kernel<T, false><<<grid, BN_BLOCK, 0, stream>>>( \
                 dyp, xp, yp, m, c, saved_mean.data_ptr<float>(), \
-                saved_rstd.data_ptr<float>(), sums); \
+                saved_rstd.data_ptr<float>(), pp); \
+        bn_reduce_partials_kernel<<<(2 * c + BN_BLOCK - 1) / BN_BLOCK, \
+                                    BN_BLOCK, 0, stream>>>(pp, grid, c, \
+                                                           sums); \
         if (relu && need_dz) \
             bn_bwd_dx_kernel<T, true, true><<<grid, BN_BLOCK, 0, stream>>>( \
                 dyp, xp, yp, dxp, dzp, m, c, \
