@@ -114,6 +114,14 @@ def make_dataset(dataset: str, n_samples: int = None, seed: int = 0,
     raise ValueError(f"unknown dataset '{dataset}'")
 
 
+def _ctc_collate(batch):
+    """Collate for variable-length CTC transcripts: stack spectrograms,
+    keep the label sequences as a list (the trainer flattens them)."""
+    xs = torch.stack([b[0] for b in batch])
+    ys = [b[1] for b in batch]
+    return xs, ys
+
+
 def make_loader(dataset, batch_size: int, rank: int = 0, world: int = 1,
                 num_workers: int = 2, drop_last: bool = True,
                 seed: int = 0) -> DataLoader:
@@ -122,7 +130,9 @@ def make_loader(dataset, batch_size: int, rank: int = 0, world: int = 1,
         sampler = DistributedSampler(dataset, num_replicas=world, rank=rank,
                                      shuffle=True, seed=seed,
                                      drop_last=drop_last)
+    collate = _ctc_collate if isinstance(dataset, SyntheticAN4) else None
     return DataLoader(dataset, batch_size=batch_size, sampler=sampler,
                       shuffle=(sampler is None), drop_last=drop_last,
                       num_workers=num_workers, pin_memory=True,
-                      persistent_workers=num_workers > 0)
+                      persistent_workers=num_workers > 0,
+                      collate_fn=collate)

@@ -71,3 +71,16 @@ def test_checkpoint_roundtrip(tmp_path):
     assert torch.allclose(opt2.compression.residuals["flat"],
                           opt.compression.residuals["flat"])
     assert torch.allclose(opt2.flat_momentum, opt.flat_momentum)
+
+
+def test_trainer_an4_ctc_path():
+    tr = DLTrainer(dnn="lstman4", dataset="an4", batch_size=4, device="cpu",
+                   amp=False, num_workers=0, n_samples=16,
+                   model_kwargs={"hidden": 16})
+    opt = gt.DistributedOptimizer(tr.optimizer, tr.net.named_parameters(),
+                                  compression=gt.compressors["gtopk"](0.01))
+    loss = tr.train(1, dist_optimizer=opt)
+    tr.update_model(dist_optimizer=opt)
+    assert loss > 0
+    m = tr.test(0)
+    assert "ctc_loss" in m
