@@ -1,0 +1,355 @@
+// Pure-HIP fused-BN kernels shared by the torch extension and the
+// standalone probe (csrc/probe_bn.hip).
+#pragma once
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#ifndef BN_BLOCK
+#define BN_BLOCK 256
+#endif
+
+// ---- 16-byte channel-chunk load/store -------------------------------------
+template <typename T> struct VecTraits;
+template <> struct VecTraits<__hip_bfloat16> {
+    static constexpr int W = 8;
+    using V = ushort4;  // 8 bf16 = 16 B, loaded as 4x u16x2
+    __device__ static void load(const __hip_bfloat16* p, float* out) {
+        const uint4 raw = *reinterpret_cast<const uint4*>(p);
+        const unsigned u[4] = {raw.x, raw.y, raw.z, raw.w};
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            out[2 * i] = __bfloat162float(
+                __ushort_as_bfloat16((unsigned short)(u[i] & 0xffff)));
+            out[2 * i + 1] = __bfloat162float(
+                __ushort_as_bfloat16((unsigned short)(u[i] >> 16)));
+        }
+    }
+    __device__ static void store(__hip_bfloat16* p, const float* in) {
+        uint4 raw;
+        unsigned u[4];
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const unsigned short lo =
+                __bfloat16_as_ushort(__float2bfloat16(in[2 * i]));
+            const unsigned short hi =
+                __bfloat16_as_ushort(__float2bfloat16(in[2 * i + 1]));
+            u[i] = (unsigned)lo | ((unsigned)hi << 16);
+        }
+        raw.x = u[0]; raw.y = u[1]; raw.z = u[2]; raw.w = u[3];
+        *reinterpret_cast<uint4*>(p) = raw;
+    }
+};
+template <> struct VecTraits<float> {
+    static constexpr int W = 4;
+    __device__ static void load(const float* p, float* out) {
+        const float4 v = *reinterpret_cast<const float4*>(p);
+        out[0] = v.x; out[1] = v.y; out[2] = v.z; out[3] = v.w;
+    }
+    __device__ static void store(float* p, const float* in) {
+        *reinterpret_cast<float4*>(p) =
+            make_float4(in[0], in[1], in[2], in[3]);
+    }
+};
+
+// ---- stats: per-channel sum / sumsq ---------------------------------------
+// Row mapping (all data kernels): each block owns a CONTIGUOUS row
+// range [r_begin, r_end) — a global round-robin row stride makes every
+// wave access open a fresh DRAM page (measured 0.25 TB/s); contiguous
+// block ranges stream at full bandwidth.  Within a block, thread t owns
+// channel-chunk t % ch_chunks and walks rows at stride BLOCK/ch_chunks
+// (requires ch_chunks <= BN_BLOCK; the fp32 C=2048 case falls back to
+// a global-stride loop).
+#define BN_ROW_RANGE(first_stmt) \
+    const long long rows_per_block = (m + gridDim.x - 1) / gridDim.x; \
+    const long long r_begin = (long long)blockIdx.x * rows_per_block; \
+    const long long r_end = \
+        (r_begin + rows_per_block < m) ? r_begin + rows_per_block : m; \
+    first_stmt
+
+template <typename T>
+__global__ __launch_bounds__(BN_BLOCK) void bn_stats_kernel(
+        const T* __restrict__ x, long long m, int c,
+        float* __restrict__ partials /* [gridDim][2C], zero-init */) {
+    constexpr int W = VecTraits<T>::W;
+    const int ch_chunks = c / W;
+    float s[W] = {0}, q[W] = {0};
+    float v[W];
+    int chunk;
+    if (ch_chunks <= BN_BLOCK) {
+        chunk = threadIdx.x % ch_chunks;
+        BN_ROW_RANGE(;)
+        const int rstep = BN_BLOCK / ch_chunks;
+        for (long long r = r_begin + threadIdx.x / ch_chunks; r < r_end;
+             r += rstep) {
+            VecTraits<T>::load(x + r * c + chunk * W, v);
+            #pragma unroll
+            for (int i = 0; i < W; ++i) {
+                s[i] += v[i];
+                q[i] += v[i] * v[i];
+            }
+        }
+    } else {  // rare fp32 large-C fallback
+        const int tid = blockIdx.x * BN_BLOCK + threadIdx.x;
+        chunk = tid % ch_chunks;
+        const long long rstride =
+            ((long long)gridDim.x * BN_BLOCK) / ch_chunks;
+        if (rstride > 0) {
+            for (long long r = tid / ch_chunks; r < m; r += rstride) {
+                VecTraits<T>::load(x + r * c + chunk * W, v);
+                #pragma unroll
+                for (int i = 0; i < W; ++i) {
+                    s[i] += v[i];
+                    q[i] += v[i] * v[i];
+                }
+            }
+        }
+    }
+    // block tree-reduction over threads sharing a chunk, then ONE
+    // non-atomic partial-row store per block (a per-channel atomicAdd
+    // builds gridDim-long same-address chains — measured 100s of µs).
+    float* mine = partials + (long long)blockIdx.x * 2 * c;
+    const int t = threadIdx.x;
+    if (ch_chunks >= BN_BLOCK) {  // unique chunk per thread in block
+        #pragma unroll
+        for (int i = 0; i < W; ++i) {
+            mine[chunk * W + i] = s[i];
+            mine[c + chunk * W + i] = q[i];
+        }
+        return;
+    }
+    __shared__ float red[BN_BLOCK];
+    #pragma unroll
+    for (int i = 0; i < W; ++i) {
+        red[t] = s[i];
+        __syncthreads();
+        for (int off = BN_BLOCK >> 1; off >= ch_chunks; off >>= 1) {
+            if (t < off) red[t] += red[t + off];
+            __syncthreads();
+        }
+        if (t < ch_chunks) mine[t * W + i] = red[t];
+        __syncthreads();
+        red[t] = q[i];
+        __syncthreads();
+        for (int off = BN_BLOCK >> 1; off >= ch_chunks; off >>= 1) {
+            if (t < off) red[t] += red[t + off];
+            __syncthreads();
+        }
+        if (t < ch_chunks) mine[c + t * W + i] = red[t];
+        __syncthreads();
+    }
+}
+
+// ---- reduce per-block partials into sums[2C] ------------------------------
+__global__ __launch_bounds__(BN_BLOCK) void bn_reduce_partials_kernel(
+        const float* __restrict__ partials, int nblocks, int c,
+        float* __restrict__ sums) {
+    for (int i = blockIdx.x * BN_BLOCK + threadIdx.x; i < 2 * c;
+         i += gridDim.x * BN_BLOCK) {
+        float acc = 0.0f;
+        for (int b = 0; b < nblocks; ++b)
+            acc += partials[(long long)b * 2 * c + i];
+        sums[i] = acc;
+    }
+}
+
+// ---- finalize: mean/rstd + running-stat update ----------------------------
+__global__ __launch_bounds__(BN_BLOCK) void bn_finalize_kernel(
+        const float* __restrict__ sums, long long m, int c, float eps,
+        float momentum, float* __restrict__ saved_mean,
+        float* __restrict__ saved_rstd, float* __restrict__ running_mean,
+        float* __restrict__ running_var) {
+    for (int i = blockIdx.x * BN_BLOCK + threadIdx.x; i < c;
+         i += gridDim.x * BN_BLOCK) {
+        const float mean = sums[i] / (float)m;
+        float var = sums[c + i] / (float)m - mean * mean;
+        if (var < 0.0f) var = 0.0f;
+        saved_mean[i] = mean;
+        saved_rstd[i] = rsqrtf(var + eps);
+        if (running_mean != nullptr) {
+            const float unbiased = m > 1 ? var * (float)m / (float)(m - 1)
+                                         : var;
+            running_mean[i] += momentum * (mean - running_mean[i]);
+            running_var[i] += momentum * (unbiased - running_var[i]);
+        }
+    }
+}
+
+// ---- forward: normalize (+add) (+relu) ------------------------------------
+template <typename T, bool RELU, bool ADD>
+__global__ __launch_bounds__(BN_BLOCK) void bn_fwd_kernel(
+        const T* __restrict__ x, const T* __restrict__ z,
+        T* __restrict__ y, long long m, int c,
+        const float* __restrict__ mean, const float* __restrict__ rstd,
+        const float* __restrict__ gamma, const float* __restrict__ beta) {
+    constexpr int W = VecTraits<T>::W;
+    const int ch_chunks = c / W;
+    long long r0, r_limit, rstep;
+    int chunk;
+    if (ch_chunks <= BN_BLOCK) {
+        BN_ROW_RANGE(;)
+        chunk = threadIdx.x % ch_chunks;
+        r0 = r_begin + threadIdx.x / ch_chunks;
+        r_limit = r_end;
+        rstep = BN_BLOCK / ch_chunks;
+    } else {
+        const int tid = blockIdx.x * BN_BLOCK + threadIdx.x;
+        chunk = tid % ch_chunks;
+        r0 = tid / ch_chunks;
+        r_limit = m;
+        rstep = ((long long)gridDim.x * BN_BLOCK) / ch_chunks;
+        if (rstep == 0) return;
+    }
+    float sc[W], sh[W];
+    #pragma unroll
+    for (int i = 0; i < W; ++i) {
+        const int ci = chunk * W + i;
+        sc[i] = gamma[ci] * rstd[ci];
+        sh[i] = beta[ci] - mean[ci] * sc[i];
+    }
+    float v[W], zr[W];
+    for (long long r = r0; r < r_limit; r += rstep) {
+        VecTraits<T>::load(x + r * c + chunk * W, v);
+        if (ADD) VecTraits<T>::load(z + r * c + chunk * W, zr);
+        #pragma unroll
+        for (int i = 0; i < W; ++i) {
+            float o = v[i] * sc[i] + sh[i];
+            if (ADD) o += zr[i];
+            if (RELU) o = o > 0.0f ? o : 0.0f;
+            v[i] = o;
+        }
+        VecTraits<T>::store(y + r * c + chunk * W, v);
+    }
+}
+
+// ---- backward stats: per-channel Σdy_eff, Σdy_eff·x̂ ----------------------
+template <typename T, bool RELU>
+__global__ __launch_bounds__(BN_BLOCK) void bn_bwd_stats_kernel(
+        const T* __restrict__ dy, const T* __restrict__ x,
+        const T* __restrict__ y, long long m, int c,
+        const float* __restrict__ mean, const float* __restrict__ rstd,
+        float* __restrict__ partials /* [gridDim][2C], zero-init */) {
+    constexpr int W = VecTraits<T>::W;
+    const int ch_chunks = c / W;
+    long long r0, r_limit, rstep;
+    int chunk;
+    if (ch_chunks <= BN_BLOCK) {
+        BN_ROW_RANGE(;)
+        chunk = threadIdx.x % ch_chunks;
+        r0 = r_begin + threadIdx.x / ch_chunks;
+        r_limit = r_end;
+        rstep = BN_BLOCK / ch_chunks;
+    } else {
+        const int tid = blockIdx.x * BN_BLOCK + threadIdx.x;
+        chunk = tid % ch_chunks;
+        r0 = tid / ch_chunks;
+        r_limit = m;
+        rstep = ((long long)gridDim.x * BN_BLOCK) / ch_chunks;
+    }
+    float mu[W], rs[W];
+    #pragma unroll
+    for (int i = 0; i < W; ++i) {
+        mu[i] = mean[chunk * W + i];
+        rs[i] = rstd[chunk * W + i];
+    }
+    float db[W] = {0}, dg[W] = {0};
+    float g[W], xv[W], yv[W];
+    if (rstep > 0) {
+        for (long long r = r0; r < r_limit; r += rstep) {
+            const long long base = r * c + chunk * W;
+            VecTraits<T>::load(dy + base, g);
+            VecTraits<T>::load(x + base, xv);
+            if (RELU) VecTraits<T>::load(y + base, yv);
+            #pragma unroll
+            for (int i = 0; i < W; ++i) {
+                const float d = (RELU && yv[i] <= 0.0f) ? 0.0f : g[i];
+                db[i] += d;
+                dg[i] += d * (xv[i] - mu[i]) * rs[i];
+            }
+        }
+    }
+    float* mine = partials + (long long)blockIdx.x * 2 * c;
+    const int t = threadIdx.x;
+    if (ch_chunks >= BN_BLOCK) {
+        #pragma unroll
+        for (int i = 0; i < W; ++i) {
+            mine[chunk * W + i] = db[i];
+            mine[c + chunk * W + i] = dg[i];
+        }
+        return;
+    }
+    __shared__ float red[BN_BLOCK];
+    #pragma unroll
+    for (int i = 0; i < W; ++i) {
+        red[t] = db[i];
+        __syncthreads();
+        for (int off = BN_BLOCK >> 1; off >= ch_chunks; off >>= 1) {
+            if (t < off) red[t] += red[t + off];
+            __syncthreads();
+        }
+        if (t < ch_chunks) mine[t * W + i] = red[t];
+        __syncthreads();
+        red[t] = dg[i];
+        __syncthreads();
+        for (int off = BN_BLOCK >> 1; off >= ch_chunks; off >>= 1) {
+            if (t < off) red[t] += red[t + off];
+            __syncthreads();
+        }
+        if (t < ch_chunks) mine[c + t * W + i] = red[t];
+        __syncthreads();
+    }
+}
+
+// ---- backward dx (+dz) ----------------------------------------------------
+template <typename T, bool RELU, bool ADD>
+__global__ __launch_bounds__(BN_BLOCK) void bn_bwd_dx_kernel(
+        const T* __restrict__ dy, const T* __restrict__ x,
+        const T* __restrict__ y, T* __restrict__ dx, T* __restrict__ dz,
+        long long m, int c, const float* __restrict__ mean,
+        const float* __restrict__ rstd, const float* __restrict__ gamma,
+        const float* __restrict__ sums /* dbeta | dgamma */) {
+    constexpr int W = VecTraits<T>::W;
+    const int ch_chunks = c / W;
+    long long r0, r_limit, rstep;
+    int chunk;
+    if (ch_chunks <= BN_BLOCK) {
+        BN_ROW_RANGE(;)
+        chunk = threadIdx.x % ch_chunks;
+        r0 = r_begin + threadIdx.x / ch_chunks;
+        r_limit = r_end;
+        rstep = BN_BLOCK / ch_chunks;
+    } else {
+        const int tid = blockIdx.x * BN_BLOCK + threadIdx.x;
+        chunk = tid % ch_chunks;
+        r0 = tid / ch_chunks;
+        r_limit = m;
+        rstep = ((long long)gridDim.x * BN_BLOCK) / ch_chunks;
+        if (rstep == 0) return;
+    }
+    const float inv_m = 1.0f / (float)m;
+    float mu[W], rs[W], gm[W], dbm[W], dgm[W];
+    #pragma unroll
+    for (int i = 0; i < W; ++i) {
+        const int ci = chunk * W + i;
+        mu[i] = mean[ci];
+        rs[i] = rstd[ci];
+        gm[i] = gamma[ci] * rs[i];
+        dbm[i] = sums[ci] * inv_m;
+        dgm[i] = sums[c + ci] * inv_m;
+    }
+    float g[W], xv[W], yv[W], dzv[W];
+    for (long long r = r0; r < r_limit; r += rstep) {
+        const long long base = r * c + chunk * W;
+        VecTraits<T>::load(dy + base, g);
+        VecTraits<T>::load(x + base, xv);
+        if (RELU) VecTraits<T>::load(y + base, yv);
+        #pragma unroll
+        for (int i = 0; i < W; ++i) {
+            const float d = (RELU && yv[i] <= 0.0f) ? 0.0f : g[i];
+            if (ADD) dzv[i] = d;
+            const float xh = (xv[i] - mu[i]) * rs[i];
+            g[i] = gm[i] * (d - dbm[i] - xh * dgm[i]);
+        }
+        VecTraits<T>::store(dx + base, g);
+        if (ADD) VecTraits<T>::store(dz + base, dzv);
+    }
+}
+
