@@ -94,9 +94,19 @@ def test_fused_bn_bf16_close_to_fp32_ref():
     assert torch.allclose(x.grad.float()[safe], x2.grad[safe], atol=5e-2), \
         (x.grad.float()[safe] - x2.grad[safe]).abs().max().item()
     assert torch.allclose(z.grad.float()[safe], z2.grad[safe], atol=5e-2)
-    # channel sums integrate over the boundary noise — keep loose
-    assert torch.allclose(fused.weight.grad, ref_bn.weight.grad, atol=0.5)
-    assert torch.allclose(fused.bias.grad, ref_bn.bias.grad, atol=0.5)
+    # dgamma/dbeta oracle honouring the FUSED kernel's own ReLU mask
+    # (channel sums would otherwise integrate the boundary flips)
+    mask = (out.detach().float() > 0)
+    dy_eff = g * mask
+    mean = x.detach().float().mean(dim=(0, 2, 3), keepdim=True)
+    var = x.detach().float().var(dim=(0, 2, 3), unbiased=False,
+                                 keepdim=True)
+    xhat = (x.detach().float() - mean) / torch.sqrt(var + fused.eps)
+    dgamma_exp = (dy_eff * xhat).sum(dim=(0, 2, 3))
+    dbeta_exp = dy_eff.sum(dim=(0, 2, 3))
+    assert torch.allclose(fused.weight.grad, dgamma_exp, atol=5e-2), \
+        (fused.weight.grad - dgamma_exp).abs().max().item()
+    assert torch.allclose(fused.bias.grad, dbeta_exp, atol=5e-2)
 
 
 def test_fused_bn_eval_mode():
