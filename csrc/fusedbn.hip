@@ -120,9 +120,14 @@ std::vector<at::Tensor> bn_fwd_train(
         bn_stats_kernel<float><<<grid, BN_BLOCK, 0, stream>>>(
             x.data_ptr<float>(), m, c, partials.data_ptr<float>());
     }
-    bn_reduce_partials_kernel<<<(2 * c + BN_BLOCK - 1) / BN_BLOCK,
-                                BN_BLOCK, 0, stream>>>(
-        partials.data_ptr<float>(), grid, c, sums);
+    {
+        const int slices = std::min(grid, std::max(1, 32768 / (2 * c)));
+        const long long work = (long long)2 * c * slices;
+        BN_OK(hipMemsetAsync(sums, 0, 2 * c * sizeof(float), stream));
+        bn_reduce_partials_kernel<<<(int)((work + BN_BLOCK - 1) / BN_BLOCK),
+                                    BN_BLOCK, 0, stream>>>(
+            partials.data_ptr<float>(), grid, c, slices, sums);
+    }
     bn_finalize_kernel<<<(c + BN_BLOCK - 1) / BN_BLOCK, BN_BLOCK, 0,
                          stream>>>(
         sums, m, c, (float)eps, (float)momentum,
@@ -196,9 +201,13 @@ std::vector<at::Tensor> bn_bwd(
             bn_bwd_stats_kernel<T, false><<<grid, BN_BLOCK, 0, stream>>>( \
                 dyp, xp, yp, m, c, saved_mean.data_ptr<float>(), \
                 saved_rstd.data_ptr<float>(), pp); \
-        bn_reduce_partials_kernel<<<(2 * c + BN_BLOCK - 1) / BN_BLOCK, \
-                                    BN_BLOCK, 0, stream>>>(pp, grid, c, \
-                                                           sums); \
+        { const int slices = std::min(grid, std::max(1, 32768 / (2 * c))); \
+          const long long work = (long long)2 * c * slices; \
+          BN_OK(hipMemsetAsync(sums, 0, 2 * c * sizeof(float), stream)); \
+          bn_reduce_partials_kernel<<<(int)((work + BN_BLOCK - 1) / \
+                                            BN_BLOCK), BN_BLOCK, 0, \
+                                      stream>>>(pp, grid, c, slices, \
+                                                sums); } \
         if (relu && need_dz) \
             bn_bwd_dx_kernel<T, true, true><<<grid, BN_BLOCK, 0, stream>>>( \
                 dyp, xp, yp, dxp, dzp, m, c, \

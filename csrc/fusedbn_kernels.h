@@ -139,15 +139,22 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_stats_kernel(
 }
 
 // ---- reduce per-block partials into sums[2C] ------------------------------
+// Parallel over (element, block-slice): 2C alone is far too few threads
+// (a 2C-thread serial loop over 2048 rows measured 474 µs — more than
+// the stats pass itself).  sums must be zeroed by the caller.
 __global__ __launch_bounds__(BN_BLOCK) void bn_reduce_partials_kernel(
         const float* __restrict__ partials, int nblocks, int c,
-        float* __restrict__ sums) {
-    for (int i = blockIdx.x * BN_BLOCK + threadIdx.x; i < 2 * c;
-         i += gridDim.x * BN_BLOCK) {
+        int slices, float* __restrict__ sums) {
+    const long long idx = (long long)blockIdx.x * BN_BLOCK + threadIdx.x;
+    const long long total = (long long)2 * c * slices;
+    for (long long t = idx; t < total;
+         t += (long long)gridDim.x * BN_BLOCK) {
+        const int i = (int)(t % (2 * c));   // consecutive threads: coalesced
+        const int s = (int)(t / (2 * c));
         float acc = 0.0f;
-        for (int b = 0; b < nblocks; ++b)
+        for (int b = s; b < nblocks; b += slices)
             acc += partials[(long long)b * 2 * c + i];
-        sums[i] = acc;
+        atomicAdd(&sums[i], acc);
     }
 }
 
