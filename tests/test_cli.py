@@ -39,3 +39,14 @@ def test_train_cli_gloo_world2(tmp_path):
          "--checkpoint-dir", str(tmp_path / "ckpt")],
         cwd=REPO, env=env, capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, (r.stdout[-1000:], r.stderr[-2000:])
+
+
+def test_collective_self_benchmark_cpu():
+    r = subprocess.run(
+        [sys.executable, "-m", "gtopkssgd_amd.parallel.allreducer",
+         "--n", "10000", "--density", "0.01", "--iters", "3",
+         "--warmup", "1", "--device", "cpu"],
+        cwd=REPO, capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stderr[-1000:]
+    out = json.loads(r.stdout.strip().splitlines()[-1])
+    assert out["collective"] == "gtopk" and out["k"] == 100
