@@ -27,7 +27,7 @@ from gtopkssgd_amd import models as zoo  # noqa: E402
 
 # model -> (builder kwargs dataset-shape, default per-GPU batch, unit)
 _CONFIGS = {
-    "resnet50": dict(shape=(3, 224, 224), classes=1000, batch=256),
+    "resnet50": dict(shape=(3, 224, 224), classes=1000, batch=512),
     "alexnet": dict(shape=(3, 224, 224), classes=1000, batch=512),
     "resnet20": dict(shape=(3, 32, 32), classes=10, batch=2048),
     "vgg16": dict(shape=(3, 32, 32), classes=10, batch=1024),
