@@ -84,6 +84,18 @@ def main():
         _prime_miopen_db()
     if world > 1:
         dist.barrier()
+        # each rank gets its own user find-db copy: 8 ranks sharing one
+        # db file serialize on MIOpen's file locks during find
+        import shutil
+        src_db = os.path.expanduser("~/.config/miopen")
+        if os.path.isdir(src_db):
+            rank_db = os.path.expanduser(f"~/.config/miopen_r{local_rank}")
+            try:
+                shutil.copytree(src_db, rank_db, dirs_exist_ok=True)
+                os.environ["MIOPEN_USER_DB_PATH"] = rank_db
+            except OSError as exc:
+                print(f"# per-rank miopen db copy failed: {exc}",
+                      flush=True)
     device = torch.device("cuda", local_rank)
     torch.manual_seed(1234)
     # MIOpen: exhaustive-find conv algo selection (amortised in warmup);

@@ -5,17 +5,17 @@
 //  * wave64 everywhere; __ballot is 64-bit; blocks are 256 threads.
 //  * All streaming kernels are grid-stride, float4-vectorized, capped at
 //    2048 blocks (Guideline 11).
-//  * Top-k select is a bucketed threshold select: one fused pass computes
-//    `residual += grad` AND an 8192-bin LDS histogram of the |value|
-//    bit-key (bits 30..18: exponent + top-5 mantissa); a single-block
-//    scan finds the threshold bucket; a second pass compacts winners with
-//    wave-aggregated atomics and zeroes them in the residual (error
-//    feedback).  Exact in count (always exactly k), ties inside the
-//    threshold bucket (a ≤3% relative-|v| band) are arbitrary — the
-//    reference's torch.topk is tie-arbitrary too.
+//  * Top-k select is a TWO-LEVEL bucketed threshold select: one fused
+//    pass computes `residual += grad` AND an 8192-bin LDS histogram of
+//    the |value| bit-key (bits 30..18); a single-block scan finds the
+//    threshold bucket; a refinement histogram over the threshold bucket
+//    (bits 17..5) makes the select exact on the top 26 magnitude bits;
+//    a final pass compacts winners through per-block LDS staging (one
+//    slot-claim atomic per chunk) and zeroes them in the residual
+//    (error feedback).  Always exactly k outputs; 26-bit ties are
+//    arbitrary, like torch.topk's.
 //  * fp32 global atomicAdd/atomicExch are native on CDNA4; scatter sizes
-//    here are O(k) ≈ 25k elements, contention-free in practice
-//    (Guideline 12: per-wave aggregation for the hot counters).
+//    here are O(k) ≈ 25k elements, contention-free in practice.
 
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
