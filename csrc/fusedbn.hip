@@ -111,7 +111,11 @@ std::vector<at::Tensor> bn_fwd_train(
     float* sums = stats_scratch.data_ptr<float>();
     const int ch = c / W;
     const int grid = bn_grid(m, ch);
-    auto partials = at::zeros({grid, 2 * c}, gamma.options());
+    // partials fully overwritten when ch<=BLOCK (tree writes every
+    // entry); the rare large-C fallback leaves gaps -> zero it then
+    auto partials = ch <= BN_BLOCK
+        ? at::empty({grid, 2 * c}, gamma.options())
+        : at::zeros({grid, 2 * c}, gamma.options());
     if (x.scalar_type() == at::kBFloat16) {
         bn_stats_kernel<__hip_bfloat16><<<grid, BN_BLOCK, 0, stream>>>(
             reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()), m, c,
@@ -182,7 +186,9 @@ std::vector<at::Tensor> bn_bwd(
     const int ch = c / W;
     TORCH_CHECK(c % W == 0 && (ch & (ch - 1)) == 0, "bad C");
     const int grid = bn_grid(m, ch);
-    auto partials = at::zeros({grid, 2 * c}, gamma.options());
+    auto partials = ch <= BN_BLOCK
+        ? at::empty({grid, 2 * c}, gamma.options())
+        : at::zeros({grid, 2 * c}, gamma.options());
     float* pp = partials.data_ptr<float>();
     TORCH_CHECK(!relu || y.has_value(), "relu backward needs y");
 
