@@ -59,6 +59,9 @@ def parse_args():
                    help="JSONL per-phase optimizer stats path (rank 0)")
     p.add_argument("--profile", action="store_true",
                    help="enable HIP-event phase timers in the optimizer")
+    p.add_argument("--torch-profile", default=None, metavar="DIR",
+                   help="capture a torch.profiler chrome trace of a few "
+                        "steps into DIR (rank 0)")
     return p.parse_args()
 
 
@@ -107,12 +110,24 @@ def main():
     logger.info("run %s: world=%d iters/epoch=%d", run_tag, world,
                 iters_per_epoch)
 
+    prof = None
+    if args.torch_profile and rank == 0:
+        from torch.profiler import (ProfilerActivity, profile, schedule,
+                                    tensorboard_trace_handler)
+        prof = profile(
+            activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA],
+            schedule=schedule(wait=2, warmup=2, active=3, repeat=1),
+            on_trace_ready=tensorboard_trace_handler(args.torch_profile))
+        prof.__enter__()
+
     for epoch in range(args.max_epochs):
         lr = trainer.adjust_learning_rate(epoch, opt)
         t0 = time.time()
         for it in range(iters_per_epoch):
             loss = trainer.train(args.nsteps_update, dist_optimizer=opt)
             trainer.update_model(dist_optimizer=opt)
+            if prof is not None:
+                prof.step()
             if rank == 0 and (it + 1) % 50 == 0:
                 wall = time.time() - t0
                 seen = (it + 1) * args.nsteps_update * args.batch_size
@@ -131,6 +146,8 @@ def main():
             trainer.save(args.checkpoint_dir, run_tag, epoch, optimizer=opt)
         trainer.timer.reset()
 
+    if prof is not None:
+        prof.__exit__(None, None, None)
     if dist.is_available() and dist.is_initialized():
         dist.destroy_process_group()
 
