@@ -50,3 +50,21 @@ def test_collective_self_benchmark_cpu():
     assert r.returncode == 0, r.stderr[-1000:]
     out = json.loads(r.stdout.strip().splitlines()[-1])
     assert out["collective"] == "gtopk" and out["k"] == 100
+
+
+def test_train_cli_resume(tmp_path):
+    base = [sys.executable, os.path.join(REPO, "train.py"),
+            "--dnn", "mnistffn", "--dataset", "mnist", "--batch-size", "8",
+            "--iters-per-epoch", "2", "--n-samples", "32",
+            "--density", "0.05", "--device", "cpu", "--no-amp",
+            "--num-workers", "0", "--run-tag", "rt",
+            "--checkpoint-dir", str(tmp_path / "ckpt")]
+    r1 = subprocess.run(base + ["--max-epochs", "1"], cwd=REPO,
+                        capture_output=True, text=True, timeout=300)
+    assert r1.returncode == 0, r1.stderr[-1000:]
+    r2 = subprocess.run(base + ["--max-epochs", "3", "--resume"], cwd=REPO,
+                        capture_output=True, text=True, timeout=300)
+    assert r2.returncode == 0, r2.stderr[-1000:]
+    assert "resumed from" in r2.stderr + r2.stdout
+    names = os.listdir(str(tmp_path / "ckpt" / "rt"))
+    assert "model-2.pth" in names

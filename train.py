@@ -54,6 +54,9 @@ def parse_args():
     p.add_argument("--norm-clip", type=float, default=None)
     p.add_argument("--checkpoint-dir", default="./checkpoints")
     p.add_argument("--run-tag", default=None)
+    p.add_argument("--resume", action="store_true",
+                   help="resume from the latest checkpoint of this run-tag "
+                        "(model + optimizer + EF residual state)")
     p.add_argument("--metrics", default=None, help="JSONL metrics path")
     p.add_argument("--stats", default=None,
                    help="JSONL per-phase optimizer stats path (rank 0)")
@@ -105,6 +108,20 @@ def main():
         compression=compressor, density=args.density,
         profile=args.profile or bool(writer), writer=writer)
 
+    start_epoch = 0
+    if args.resume:
+        from gtopkssgd_amd.checkpoint import latest_checkpoint, \
+            load_checkpoint
+        path = latest_checkpoint(args.checkpoint_dir, run_tag)
+        if path:
+            start_epoch = load_checkpoint(path, trainer.net, opt,
+                                          map_location=str(trainer.device)
+                                          ) + 1
+            logger.info("resumed from %s (next epoch %d)", path,
+                        start_epoch)
+        else:
+            logger.info("no checkpoint found for %s — fresh start", run_tag)
+
     iters_per_epoch = args.iters_per_epoch or max(
         1, len(trainer.trainloader) // args.nsteps_update)
     logger.info("run %s: world=%d iters/epoch=%d", run_tag, world,
@@ -120,7 +137,7 @@ def main():
             on_trace_ready=tensorboard_trace_handler(args.torch_profile))
         prof.__enter__()
 
-    for epoch in range(args.max_epochs):
+    for epoch in range(start_epoch, args.max_epochs):
         lr = trainer.adjust_learning_rate(epoch, opt)
         t0 = time.time()
         for it in range(iters_per_epoch):
