@@ -60,8 +60,8 @@ static int bn_grid(long long m, int ch_chunks) {
 
 template <typename T>
 static void bn_fwd_t(const at::Tensor& x, const c10::optional<at::Tensor>& z,
-                     at::Tensor& y, long long m, int c,
-                     const at::Tensor& mean, const at::Tensor& rstd,
+                     at::Tensor& y, unsigned char* maskp, long long m,
+                     int c, const at::Tensor& mean, const at::Tensor& rstd,
                      const at::Tensor& gamma, const at::Tensor& beta,
                      bool relu, hipStream_t stream) {
     const int ch = c / VecTraits<T>::W;
@@ -72,7 +72,7 @@ static void bn_fwd_t(const at::Tensor& x, const c10::optional<at::Tensor>& z,
     T* yp = reinterpret_cast<T*>(y.data_ptr());
     #define LAUNCH_FWD(R, A) \
         bn_fwd_kernel<T, R, A><<<grid, BN_BLOCK, 0, stream>>>( \
-            xp, zp, yp, m, c, mean.data_ptr<float>(), \
+            xp, zp, yp, maskp, m, c, mean.data_ptr<float>(), \
             rstd.data_ptr<float>(), gamma.data_ptr<float>(), \
             beta.data_ptr<float>())
     if (relu && zp) LAUNCH_FWD(true, true);
@@ -138,13 +138,21 @@ std::vector<at::Tensor> bn_fwd_train(
         saved_mean.data_ptr<float>(), saved_rstd.data_ptr<float>(),
         running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
         running_var.defined() ? running_var.data_ptr<float>() : nullptr);
+    at::Tensor mask;
+    unsigned char* maskp = nullptr;
+    if (relu) {
+        mask = at::empty({m * ch},
+                         x.options().dtype(at::kByte));
+        maskp = mask.data_ptr<unsigned char>();
+    }
     if (x.scalar_type() == at::kBFloat16)
-        bn_fwd_t<__hip_bfloat16>(x, z, y, m, c, saved_mean, saved_rstd,
-                                 gamma, beta, relu, stream);
+        bn_fwd_t<__hip_bfloat16>(x, z, y, maskp, m, c, saved_mean,
+                                 saved_rstd, gamma, beta, relu, stream);
     else
-        bn_fwd_t<float>(x, z, y, m, c, saved_mean, saved_rstd, gamma,
-                        beta, relu, stream);
+        bn_fwd_t<float>(x, z, y, maskp, m, c, saved_mean, saved_rstd,
+                        gamma, beta, relu, stream);
     BN_OK(hipGetLastError());
+    if (relu) return {y, saved_mean, saved_rstd, mask};
     return {y, saved_mean, saved_rstd};
 }
 
@@ -160,18 +168,18 @@ at::Tensor bn_fwd_eval(at::Tensor x, c10::optional<at::Tensor> z,
     auto y = at::empty_like(x);
     auto rstd = at::rsqrt(running_var + eps);
     if (x.scalar_type() == at::kBFloat16)
-        bn_fwd_t<__hip_bfloat16>(x, z, y, m, c, running_mean, rstd,
-                                 gamma, beta, relu, stream);
+        bn_fwd_t<__hip_bfloat16>(x, z, y, nullptr, m, c, running_mean,
+                                 rstd, gamma, beta, relu, stream);
     else
-        bn_fwd_t<float>(x, z, y, m, c, running_mean, rstd, gamma, beta,
-                        relu, stream);
+        bn_fwd_t<float>(x, z, y, nullptr, m, c, running_mean, rstd,
+                        gamma, beta, relu, stream);
     BN_OK(hipGetLastError());
     return y;
 }
 
 // backward: returns (dx, dgamma, dbeta, dz?)
 std::vector<at::Tensor> bn_bwd(
-        at::Tensor dy, at::Tensor x, c10::optional<at::Tensor> y,
+        at::Tensor dy, at::Tensor x, c10::optional<at::Tensor> mask,
         at::Tensor gamma, at::Tensor saved_mean, at::Tensor saved_rstd,
         bool relu, bool need_dz, at::Tensor stats_scratch) {
     const int c = (int)x.size(1);
@@ -190,11 +198,14 @@ std::vector<at::Tensor> bn_bwd(
         ? at::empty({grid, 2 * c}, gamma.options())
         : at::zeros({grid, 2 * c}, gamma.options());
     float* pp = partials.data_ptr<float>();
-    TORCH_CHECK(!relu || y.has_value(), "relu backward needs y");
+    TORCH_CHECK(!relu || (mask.has_value() &&
+                          mask->scalar_type() == at::kByte &&
+                          mask->numel() >= m * ch),
+                "relu backward needs the forward's packed mask");
 
     #define DISPATCH_BWD(T, TPTR) do { \
-        const T* yp = y.has_value() \
-            ? reinterpret_cast<const T*>(y->data_ptr()) : nullptr; \
+        const unsigned char* yp = mask.has_value() \
+            ? mask->data_ptr<unsigned char>() : nullptr; \
         const T* dyp = reinterpret_cast<const T*>(dyc.data_ptr()); \
         const T* xp = reinterpret_cast<const T*>(x.data_ptr()); \
         T* dxp = reinterpret_cast<T*>(dx.data_ptr()); \

@@ -45,6 +45,7 @@ int main() {
         const int c = sh.c;
         const long long n = m * c;
         bf16 *x, *z, *y, *dy, *dx, *dz;
+        unsigned char *mask;
         float *params, *partials, *sums;
         CHECK(hipMalloc(&x, n * 2));
         CHECK(hipMalloc(&z, n * 2));
@@ -52,6 +53,8 @@ int main() {
         CHECK(hipMalloc(&dy, n * 2));
         CHECK(hipMalloc(&dx, n * 2));
         CHECK(hipMalloc(&dz, n * 2));
+        CHECK(hipMalloc(&mask, n / 8 + 16));
+        CHECK(hipMemset(mask, 0xaa, n / 8 + 16));
         CHECK(hipMalloc(&params, 8 * c * 4));
         CHECK(hipMemset(x, 0x3c, n * 2));  // ~0.0117 bf16 pattern
         CHECK(hipMemset(z, 0x3c, n * 2));
@@ -91,14 +94,15 @@ int main() {
             bn_stats_kernel<bf16><<<grid, BN_BLOCK>>>(x, m, c, partials); });
         bench("fwd(add,relu)", n * 2 * 3.0, [&] {
             bn_fwd_kernel<bf16, true, true><<<grid, BN_BLOCK>>>(
-                x, z, y, m, c, mean, rstd, gamma, beta); });
-        bench("bwd_stats", n * 2 * 3.0, [&] {
+                x, z, y, mask, m, c, mean, rstd, gamma, beta); });
+        bench("bwd_stats", n * 2 * 2.0 + n / 8.0, [&] {
             bn_bwd_stats_kernel<bf16, true><<<grid, BN_BLOCK>>>(
-                dy, x, y, m, c, mean, rstd, partials); });
-        bench("bwd_dx", n * 2 * 5.0, [&] {
+                dy, x, mask, m, c, mean, rstd, partials); });
+        bench("bwd_dx", n * 2 * 4.0 + n / 8.0, [&] {
             bn_bwd_dx_kernel<bf16, true, true><<<grid, BN_BLOCK>>>(
-                dy, x, y, dx, dz, m, c, mean, rstd, gamma, sums); });
+                dy, x, mask, dx, dz, m, c, mean, rstd, gamma, sums); });
         hipFree(x); hipFree(z); hipFree(y); hipFree(dy); hipFree(dx);
+        hipFree(mask);
         hipFree(dz); hipFree(params); hipFree(partials); hipFree(sums);
     }
     return 0;

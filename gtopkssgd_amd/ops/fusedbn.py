@@ -40,20 +40,23 @@ class _FusedBNFunc(torch.autograd.Function):
     def forward(ctx, x, z, weight, bias, running_mean, running_var,
                 momentum, eps, relu):
         scratch = _scratch_for(x.device)
-        y, mean, rstd = _ext().bn_fwd_train(
+        outs = _ext().bn_fwd_train(
             x, z, weight, bias, running_mean, running_var, momentum, eps,
             relu, scratch)
-        ctx.save_for_backward(x, y, weight, mean, rstd)
+        y, mean, rstd = outs[0], outs[1], outs[2]
+        mask = outs[3] if relu else torch.empty(
+            0, dtype=torch.uint8, device=x.device)
+        ctx.save_for_backward(x, mask, weight, mean, rstd)
         ctx.relu = relu
         ctx.has_z = z is not None
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, y, weight, mean, rstd = ctx.saved_tensors
+        x, mask, weight, mean, rstd = ctx.saved_tensors
         scratch = _scratch_for(x.device)
-        outs = _ext().bn_bwd(dy, x, y if ctx.relu else None, weight, mean,
-                             rstd, ctx.relu, ctx.has_z, scratch)
+        outs = _ext().bn_bwd(dy, x, mask if ctx.relu else None, weight,
+                             mean, rstd, ctx.relu, ctx.has_z, scratch)
         dx, dgamma, dbeta = outs[0], outs[1], outs[2]
         dz = outs[3] if ctx.has_z else None
         return dx, dz, dgamma, dbeta, None, None, None, None, None
