@@ -69,6 +69,8 @@ def gtopk_sparse_allreduce(
     rank_ = dist.get_rank(group)
 
     ridx, rval = _pack_views(recv_buf, k)
+    # candidate scratch reused across rounds: [0:k)=own idx, [k:2k)=recv
+    cand_idx = torch.empty(2 * k, dtype=torch.int32, device=device)
     group_size = 1
     alive = True
     while group_size < world:
@@ -80,7 +82,8 @@ def gtopk_sparse_allreduce(
                     # merge: union-sum then re-select k of ≤2k candidates
                     ops.scatter_add_(dense_scratch, sidx, sval)
                     ops.scatter_add_(dense_scratch, ridx, rval)
-                    cand_idx = torch.cat([sidx, ridx])
+                    cand_idx[:k].copy_(sidx)
+                    cand_idx[k:].copy_(ridx)
                     cand_val = ops.gather_zero(dense_scratch, cand_idx)
                     mval, midx = ops.topk_among(cand_val, cand_idx, k)
                     sidx.copy_(midx)

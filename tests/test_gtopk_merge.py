@@ -121,3 +121,11 @@ def test_grad_reducer_wrapper():
     val, idx, _ = _local_sparse(0)
     gval, gidx = r.gtopk(val, idx)
     assert gval.numel() == K
+
+
+def test_gtopk_world8():
+    run_in_procs(_worker_gtopk, 8, timeout=240)
+
+
+def test_gtopk_world5_nonpow2():
+    run_in_procs(_worker_gtopk, 5, timeout=240)
