@@ -31,3 +31,4 @@ from .parallel.allreducer import (  # noqa: F401
     topk_sparse_allreduce,
     dense_allreduce,
 )
+from .ops.fusedbn import FusedBNAct  # noqa: F401
