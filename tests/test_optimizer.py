@@ -245,3 +245,80 @@ def _worker_overlap_with_accumulation(rank, world):
 
 def test_dense_overlap_accumulation_world2():
     run_in_procs(_worker_overlap_with_accumulation, 2)
+
+
+def test_adam_inner_optimizer_generic_path():
+    """Reference API wraps ANY torch optimizer — Adam uses the
+    densify-into-.grad + inner step() fallback."""
+    torch.manual_seed(0)
+    model = _make_model()
+    inner = torch.optim.Adam(model.parameters(), lr=1e-3)
+    opt = gt.DistributedOptimizer(inner, model.named_parameters(),
+                                  compression=gt.compressors["gtopk"](0.1))
+    assert not opt._fused
+    crit = nn.CrossEntropyLoss()
+    p0 = torch.cat([p.data.view(-1) for p in model.parameters()]).clone()
+    for s in range(3):
+        x, y = _batch(5000 + s)
+        opt.zero_grad()
+        crit(model(x), y).backward()
+        opt.step()
+    p1 = torch.cat([p.data.view(-1) for p in model.parameters()])
+    assert not torch.allclose(p0, p1)
+    assert torch.isfinite(p1).all()
+
+
+def test_multiple_param_groups_generic_path():
+    torch.manual_seed(0)
+    model = _make_model()
+    inner = torch.optim.SGD([
+        {"params": model.fc1.parameters(), "lr": 0.1},
+        {"params": model.fc2.parameters(), "lr": 0.01},
+    ], momentum=0.9)
+    opt = gt.DistributedOptimizer(inner, model.named_parameters(),
+                                  compression=gt.compressors["gtopk"](0.1))
+    assert not opt._fused  # >1 group -> inner step
+    crit = nn.CrossEntropyLoss()
+    x, y = _batch(6000)
+    opt.zero_grad()
+    crit(model(x), y).backward()
+    opt.step()
+    assert opt.train_iter == 1
+
+
+def test_err_callback_fires():
+    model = _make_model()
+    inner = torch.optim.SGD(model.parameters(), lr=0.1)
+    seen = []
+    opt = gt.DistributedOptimizer(inner, model.named_parameters(),
+                                  compression=gt.compressors["gtopk"](0.1),
+                                  err_callback=seen.append)
+    # corrupt state to force a failure inside synchronize
+    opt.compression = None
+    opt.zero_grad()
+    p = inner.param_groups[0]["params"][0]
+    (p ** 2).sum().backward()
+    try:
+        opt.step()
+    except Exception:
+        pass
+    assert len(seen) == 1
+
+
+def _worker_lstm_gloo(rank, world):
+    from gtopkssgd_amd.trainer import DLTrainer
+    tr = DLTrainer(rank=rank, size=world, dnn="lstm", dataset="ptb",
+                   batch_size=4, device="cpu", amp=False, num_workers=0,
+                   n_samples=32, model_kwargs={"vocab_size": 80,
+                                               "hidden": 16})
+    opt = gt.DistributedOptimizer(tr.optimizer, tr.net.named_parameters(),
+                                  compression=gt.compressors["gtopk"](0.01))
+    for _ in range(2):
+        tr.train(1, dist_optimizer=opt)
+        tr.update_model(dist_optimizer=opt)
+    m = tr.test(0)
+    assert "perplexity" in m and m["perplexity"] > 0
+
+
+def test_lstm_gloo_world2():
+    run_in_procs(_worker_lstm_gloo, 2, timeout=240)
