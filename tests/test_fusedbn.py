@@ -55,3 +55,28 @@ def test_eval_mode_cpu():
     x = torch.randn(2, 8, 4, 4)
     out = fused(x)
     assert (out >= 0).all()
+
+
+def test_no_running_stats_fallback():
+    fused = FusedBNAct(8, relu=True, track_running_stats=False)
+    x = torch.randn(2, 8, 4, 4, requires_grad=True)
+    out = fused(x)
+    out.sum().backward()
+    assert (out >= 0).all() and x.grad is not None
+
+
+def test_momentum_none_fallback():
+    fused = FusedBNAct(8, momentum=None)
+    x = torch.randn(2, 8, 4, 4)
+    out = fused(x)
+    assert out.shape == x.shape
+
+
+def test_non_pow2_channels_fall_back(monkeypatch):
+    # 24 channels: 24/8=3 not a power of two -> fused path must decline
+    fused = FusedBNAct(24, relu=True)
+    x = torch.randn(2, 24, 4, 4)
+    assert not fused._fusable(x.cuda() if torch.cuda.is_available() else x,
+                              None) or not torch.cuda.is_available()
+    out = fused(x)
+    assert out.shape == x.shape

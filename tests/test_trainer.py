@@ -84,3 +84,24 @@ def test_trainer_an4_ctc_path():
     assert loss > 0
     m = tr.test(0)
     assert "ctc_loss" in m
+
+
+def test_lr_schedule_all_datasets():
+    for ds, dnn in [("mnist", "mnistffn"), ("ptb", "lstm")]:
+        kw = {"model_kwargs": {"vocab_size": 50, "hidden": 16}} \
+            if dnn == "lstm" else {}
+        tr = DLTrainer(dnn=dnn, dataset=ds, batch_size=4, device="cpu",
+                       amp=False, num_workers=0, n_samples=16, **kw)
+        assert tr.adjust_learning_rate(0) == tr.base_lr
+
+
+def test_latest_checkpoint_empty_dir(tmp_path):
+    assert latest_checkpoint(str(tmp_path), "nothing") is None
+
+
+def test_get_acc_resets():
+    tr = _tiny_trainer()
+    tr.train(1)
+    a = tr.get_acc()
+    assert a > 0
+    assert tr._loss_count == 0
