@@ -129,3 +129,34 @@ def test_gtopk_world8():
 
 def test_gtopk_world5_nonpow2():
     run_in_procs(_worker_gtopk, 5, timeout=240)
+
+
+def test_halving_schedule_is_deadlock_free():
+    """Simulate the send/recv schedule for P up to 64: every send must
+    have exactly one matching recv in the same round, each rank is in
+    at most one operation per round, and rank 0 ends with the result."""
+    for world in list(range(2, 20)) + [32, 33, 64]:
+        alive = set(range(world))
+        group_size = 1
+        while group_size < world:
+            sends, recvs = {}, {}
+            for r in range(world):
+                if r not in alive:
+                    continue
+                if r % (2 * group_size) < group_size:
+                    peer = r + group_size
+                    if peer < world:
+                        recvs[r] = peer
+                else:
+                    sends[r] = r - group_size
+            for sender, dst in sends.items():
+                assert recvs.get(dst) == sender, (world, group_size,
+                                                  sender, dst)
+                alive.discard(sender)
+            for receiver, src in recvs.items():
+                if src in sends:
+                    assert sends[src] == receiver
+                # recv only posted when the peer exists and is alive
+                assert src in sends or src >= world or src not in alive
+            group_size *= 2
+        assert 0 in alive

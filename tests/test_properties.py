@@ -3,11 +3,13 @@ import hypothesis.strategies as st
 import torch
 from hypothesis import given, settings
 
+# derandomize: the driver runs this suite as a gate — no random flakes
+
 from gtopkssgd_amd import ops
 from gtopkssgd_amd.compression import TopKCompressor
 
 
-@settings(max_examples=30, deadline=None)
+@settings(max_examples=30, deadline=None, derandomize=True)
 @given(n=st.integers(4, 500), seed=st.integers(0, 10_000),
        density=st.floats(0.01, 1.0))
 def test_ef_compress_invariants(n, seed, density):
@@ -31,7 +33,7 @@ def test_ef_compress_invariants(n, seed, density):
     assert torch.count_nonzero(res[i.long()]) == 0
 
 
-@settings(max_examples=20, deadline=None)
+@settings(max_examples=20, deadline=None, derandomize=True)
 @given(n=st.integers(8, 300), m=st.integers(1, 40),
        seed=st.integers(0, 10_000))
 def test_gather_zero_scatter_add_roundtrip(n, m, seed):
@@ -47,7 +49,7 @@ def test_gather_zero_scatter_add_roundtrip(n, m, seed):
     assert torch.count_nonzero(dense) == 0
 
 
-@settings(max_examples=20, deadline=None)
+@settings(max_examples=20, deadline=None, derandomize=True)
 @given(seed=st.integers(0, 10_000), steps=st.integers(1, 6))
 def test_residual_mass_conservation_over_steps(seed, steps):
     g = torch.Generator().manual_seed(seed)
@@ -63,7 +65,7 @@ def test_residual_mass_conservation_over_steps(seed, steps):
     assert torch.allclose(emitted + comp.residuals["p"], total, atol=1e-5)
 
 
-@settings(max_examples=20, deadline=None)
+@settings(max_examples=20, deadline=None, derandomize=True)
 @given(n=st.integers(10, 200), seed=st.integers(0, 10_000))
 def test_sgd_step_equals_torch_over_random_configs(n, seed):
     g = torch.Generator().manual_seed(seed)
