@@ -109,11 +109,11 @@ class FusedBNAct(nn.BatchNorm2d):
                     x, residual, self.weight, self.bias,
                     self.running_mean, self.running_var, self.eps,
                     self.relu)
-        # eager reference path (CPU / non-conforming shapes / eval+grad)
-        out = F.batch_norm(
-            x, self.running_mean, self.running_var, self.weight, self.bias,
-            self.training or not self.track_running_stats,
-            self.momentum if self.momentum is not None else 0.0, self.eps)
+        # eager reference path (CPU / non-conforming shapes / eval+grad):
+        # delegate the BN part to nn.BatchNorm2d.forward so momentum=None
+        # (cumulative moving average) and num_batches_tracked semantics
+        # match exactly, then apply the add/ReLU epilogue.
+        out = super().forward(x)
         if residual is not None:
             out = out + residual
         if self.relu:

@@ -203,18 +203,24 @@ class DLTrainer:
         return lr
 
     # ------------------------------------------------------------------
-    @staticmethod
-    def _sync_metrics(metrics: dict) -> dict:
-        """C6: average scalar metrics across ranks when distributed."""
+    def _sync_metrics(self, metrics: dict) -> dict:
+        """C6: average scalar metrics across ranks when distributed.
+
+        The scalar tensor lives on the backend's device: RCCL (backend
+        "nccl") can only reduce GPU-resident tensors, gloo reduces CPU
+        tensors — all-reducing a CPU tensor under NCCL raises.
+        """
         import torch.distributed as dist
         if not (dist.is_available() and dist.is_initialized()) or \
                 dist.get_world_size() == 1:
             return metrics
+        backend = dist.get_backend()
+        dev = self.device if backend == "nccl" else torch.device("cpu")
         keys = sorted(metrics)
-        t = torch.tensor([float(metrics[k]) for k in keys])
+        t = torch.tensor([float(metrics[k]) for k in keys], device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.SUM)
         t /= dist.get_world_size()
-        return {k: float(v) for k, v in zip(keys, t)}
+        return {k: float(v) for k, v in zip(keys, t.cpu())}
 
     @torch.no_grad()
     def test(self, epoch: int = 0):
@@ -263,9 +269,10 @@ class DLTrainer:
 
     # ------------------------------------------------------------------
     def save(self, prefix: str, run_tag: str, epoch: int, optimizer=None):
-        if self.rank == 0:
-            return save_checkpoint(prefix, run_tag, epoch, self.net,
-                                   optimizer)
+        """Call from every rank: rank 0 writes model+optimizer, every
+        rank writes its own EF-residual sidecar."""
+        return save_checkpoint(prefix, run_tag, epoch, self.net,
+                               optimizer, rank=self.rank)
 
     def get_acc(self):
         """Reference parity: running average training loss since the

@@ -66,10 +66,20 @@ def test_no_running_stats_fallback():
 
 
 def test_momentum_none_fallback():
+    """momentum=None must follow nn.BatchNorm2d's cumulative-moving-
+    average running stats (and bump num_batches_tracked), not freeze."""
+    torch.manual_seed(1)
     fused = FusedBNAct(8, momentum=None)
-    x = torch.randn(2, 8, 4, 4)
-    out = fused(x)
-    assert out.shape == x.shape
+    ref = nn.BatchNorm2d(8, momentum=None)
+    ref.load_state_dict(fused.state_dict())
+    for _ in range(3):
+        x = torch.randn(2, 8, 4, 4)
+        out = fused(x)
+        out_ref = ref(x)
+        assert torch.allclose(out, out_ref, atol=1e-6)
+    assert torch.allclose(fused.running_mean, ref.running_mean, atol=1e-6)
+    assert torch.allclose(fused.running_var, ref.running_var, atol=1e-6)
+    assert int(fused.num_batches_tracked) == int(ref.num_batches_tracked) == 3
 
 
 def test_non_pow2_channels_fall_back(monkeypatch):

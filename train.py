@@ -76,7 +76,10 @@ def init_dist(args):
         rank = dist.get_rank()
         world = dist.get_world_size()
         if use_gpu:
-            torch.cuda.set_device(rank % torch.cuda.device_count())
+            # torchrun exports LOCAL_RANK; fall back to packed layout
+            local = int(os.environ.get(
+                "LOCAL_RANK", rank % torch.cuda.device_count()))
+            torch.cuda.set_device(local)
         return rank, world
     return 0, 1
 
@@ -115,8 +118,8 @@ def main():
         path = latest_checkpoint(args.checkpoint_dir, run_tag)
         if path:
             start_epoch = load_checkpoint(path, trainer.net, opt,
-                                          map_location=str(trainer.device)
-                                          ) + 1
+                                          map_location=str(trainer.device),
+                                          rank=rank) + 1
             logger.info("resumed from %s (next epoch %d)", path,
                         start_epoch)
         else:
@@ -160,7 +163,8 @@ def main():
             if metrics:
                 metrics.log(epoch=epoch, lr=lr, samples_per_sec=sps,
                             world=world, **results)
-            trainer.save(args.checkpoint_dir, run_tag, epoch, optimizer=opt)
+        # every rank: rank 0 writes model+optimizer, each rank its EF sidecar
+        trainer.save(args.checkpoint_dir, run_tag, epoch, optimizer=opt)
         trainer.timer.reset()
 
     if prof is not None:
