@@ -83,6 +83,52 @@ __global__ __launch_bounds__(BLOCK) void ef_accum_hist_kernel(
     }
 }
 
+// Range variant of ef_accum_hist for backward-overlap: residual[s:e) +=
+// grad[s:e) and level-1 histogram of the updated values.  Buckets are
+// cut at parameter boundaries (arbitrary offsets), so the float4 body
+// is bracketed by scalar head/tail to honour 16 B alignment.  Additive
+// across disjoint ranges: the union of bucket calls over [0, n) equals
+// one ef_accum_hist_kernel call.
+__global__ __launch_bounds__(BLOCK) void ef_accum_hist_range_kernel(
+        float* __restrict__ residual, const float* __restrict__ grad,
+        long long start, long long end, unsigned* __restrict__ ghist) {
+    __shared__ unsigned lhist[HIST_BINS];
+    for (int j = threadIdx.x; j < HIST_BINS; j += BLOCK) lhist[j] = 0;
+    __syncthreads();
+    long long astart = (start + 3) & ~3LL;   // first float4-aligned elem
+    long long aend = end & ~3LL;             // end of float4 body
+    if (aend < astart) { astart = end; aend = end; }  // tiny range
+    const long long stride = (long long)gridDim.x * BLOCK;
+    const long long tid = (long long)blockIdx.x * BLOCK + threadIdx.x;
+    for (long long i = start + tid; i < astart; i += stride) {
+        const float r = residual[i] + grad[i];
+        residual[i] = r;
+        atomicAdd(&lhist[mag_key(r)], 1u);
+    }
+    for (long long i = aend + tid; i < end; i += stride) {
+        const float r = residual[i] + grad[i];
+        residual[i] = r;
+        atomicAdd(&lhist[mag_key(r)], 1u);
+    }
+    float4* r4 = reinterpret_cast<float4*>(residual);
+    const float4* g4 = reinterpret_cast<const float4*>(grad);
+    for (long long i = (astart >> 2) + tid; i < (aend >> 2); i += stride) {
+        float4 r = r4[i];
+        const float4 g = g4[i];
+        r.x += g.x; r.y += g.y; r.z += g.z; r.w += g.w;
+        r4[i] = r;
+        atomicAdd(&lhist[mag_key(r.x)], 1u);
+        atomicAdd(&lhist[mag_key(r.y)], 1u);
+        atomicAdd(&lhist[mag_key(r.z)], 1u);
+        atomicAdd(&lhist[mag_key(r.w)], 1u);
+    }
+    __syncthreads();
+    for (int j = threadIdx.x; j < HIST_BINS; j += BLOCK) {
+        unsigned c = lhist[j];
+        if (c) atomicAdd(&ghist[j], c);
+    }
+}
+
 // histogram-only variant for topk_among (candidate values)
 __global__ __launch_bounds__(BLOCK) void hist_vals_kernel(
         const float* __restrict__ vals, long long n,
@@ -293,6 +339,53 @@ __global__ __launch_bounds__(BLOCK) void compact_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Fused merge-round kernels (C1 bookkeeping, K4+K5+K8 batched).
+// A merge round on the receiver is:  dense[own ∪ recv] += vals;
+// candidates = gather-and-zero(dense, own ∪ recv idx); top-k of the
+// candidates.  The wire format packs idx at buf[0:k] and fp32 values
+// bitcast at buf[k:2k], so both kernels take the raw 2k-int32 buffers.
+// Messages are ~2k ≈ 51k items: latency-bound, so the point is FEWER
+// LAUNCHES per round (7 total incl. select), not per-kernel bandwidth.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(BLOCK) void merge_scatter2_kernel(
+        float* __restrict__ dense, const int* __restrict__ abuf,
+        const int* __restrict__ bbuf, int k) {
+    const long long stride = (long long)gridDim.x * BLOCK;
+    const float* aval = reinterpret_cast<const float*>(abuf + k);
+    const float* bval = reinterpret_cast<const float*>(bbuf + k);
+    for (long long i = (long long)blockIdx.x * BLOCK + threadIdx.x; i < 2 * k;
+         i += stride) {
+        if (i < k) atomicAdd(&dense[abuf[i]], aval[i]);
+        else       atomicAdd(&dense[bbuf[i - k]], bval[i - k]);
+    }
+}
+
+// gather-and-zero the 2k candidates, write the (val, idx) candidate
+// arrays AND the level-1 histogram in the same pass
+__global__ __launch_bounds__(BLOCK) void merge_gather_hist_kernel(
+        float* __restrict__ dense, const int* __restrict__ abuf,
+        const int* __restrict__ bbuf, int k, float* __restrict__ cand_val,
+        int* __restrict__ cand_idx, unsigned* __restrict__ ghist) {
+    __shared__ unsigned lhist[HIST_BINS];
+    for (int j = threadIdx.x; j < HIST_BINS; j += BLOCK) lhist[j] = 0;
+    __syncthreads();
+    const long long stride = (long long)gridDim.x * BLOCK;
+    for (long long i = (long long)blockIdx.x * BLOCK + threadIdx.x; i < 2 * k;
+         i += stride) {
+        const int idx = i < k ? abuf[i] : bbuf[i - k];
+        const float v = atomicExch(&dense[idx], 0.0f);  // dedup: dup reads 0
+        cand_val[i] = v;
+        cand_idx[i] = idx;
+        atomicAdd(&lhist[mag_key(v)], 1u);
+    }
+    __syncthreads();
+    for (int j = threadIdx.x; j < HIST_BINS; j += BLOCK) {
+        unsigned c = lhist[j];
+        if (c) atomicAdd(&ghist[j], c);
+    }
+}
+
+// ---------------------------------------------------------------------------
 // K4: scatter-add / gather-zero (merge round primitives)
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(BLOCK) void scatter_add_kernel(
@@ -449,6 +542,95 @@ void ef_compress(at::Tensor residual, at::Tensor grad, int64_t k,
     check_launch("gtopk");
 }
 
+// --- backward-overlap select: zero hist, accumulate per-bucket ranges
+// during backward, finish (scan+refine+compact) at synchronize() ---
+
+void ef_hist_zero(at::Tensor hist) {
+    CHECK_I32(hist);
+    TORCH_CHECK(hist.numel() >= 2 * HIST_BINS);
+    HIP_OK(hipMemsetAsync(hist.data_ptr<int>(), 0,
+                          2 * HIST_BINS * sizeof(int), cur_stream()));
+}
+
+void ef_accum_hist_range(at::Tensor residual, at::Tensor grad,
+                         int64_t start, int64_t end, at::Tensor hist) {
+    CHECK_F32(residual); CHECK_F32(grad); CHECK_I32(hist);
+    const long long n = residual.numel();
+    TORCH_CHECK(grad.numel() == n && 0 <= start && start <= end && end <= n);
+    if (start == end) return;
+    const int blocks = stream_blocks((end - start + 3) / 4);
+    ef_accum_hist_range_kernel<<<blocks, BLOCK, 0, cur_stream()>>>(
+        residual.data_ptr<float>(), grad.data_ptr<float>(), start, end,
+        (unsigned*)hist.data_ptr<int>());
+    check_launch("gtopk");
+}
+
+void ef_compress_finish(at::Tensor residual, int64_t k, at::Tensor hist,
+                        at::Tensor info, at::Tensor counters,
+                        at::Tensor out_val, at::Tensor out_idx) {
+    CHECK_F32(residual); CHECK_F32(out_val);
+    CHECK_I32(hist); CHECK_I32(info); CHECK_I32(counters); CHECK_I32(out_idx);
+    const long long n = residual.numel();
+    TORCH_CHECK(k > 0 && k <= n && hist.numel() >= 2 * HIST_BINS &&
+                out_val.numel() >= k && out_idx.numel() >= k);
+    auto stream = cur_stream();
+    unsigned* h1 = (unsigned*)hist.data_ptr<int>();
+    unsigned* h2 = h1 + HIST_BINS;
+    scan_threshold_kernel<0><<<1, BLOCK, 0, stream>>>(
+        h1, (int)k, info.data_ptr<int>(), counters.data_ptr<int>());
+    refine_hist_kernel<<<stream_blocks(n), BLOCK, 0, stream>>>(
+        residual.data_ptr<float>(), n, info.data_ptr<int>(), h2);
+    scan_threshold_kernel<1><<<1, BLOCK, 0, stream>>>(
+        h2, (int)k, info.data_ptr<int>(), counters.data_ptr<int>());
+    compact_kernel<true, false><<<stream_blocks(n), BLOCK, 0, stream>>>(
+        residual.data_ptr<float>(), nullptr, n, info.data_ptr<int>(),
+        counters.data_ptr<int>(), out_val.data_ptr<float>(),
+        out_idx.data_ptr<int>(), (int)k);
+    check_launch("gtopk");
+}
+
+// One fused merge round: dense[a ∪ b] += vals; gather-zero candidates
+// (+ level-1 hist); two-level select of k among 2k.  The merged top-k
+// is written into out_val/out_idx — pass views of the send buffer so
+// the result is wire-ready with no extra copy (K8).
+void merge_round(at::Tensor dense, at::Tensor send_buf, at::Tensor recv_buf,
+                 int64_t k, at::Tensor hist, at::Tensor info,
+                 at::Tensor counters, at::Tensor cand_val,
+                 at::Tensor cand_idx, at::Tensor out_val, at::Tensor out_idx) {
+    CHECK_F32(dense); CHECK_I32(send_buf); CHECK_I32(recv_buf);
+    CHECK_I32(hist); CHECK_I32(info); CHECK_I32(counters);
+    CHECK_F32(cand_val); CHECK_I32(cand_idx); CHECK_F32(out_val);
+    CHECK_I32(out_idx);
+    TORCH_CHECK(send_buf.numel() >= 2 * k && recv_buf.numel() >= 2 * k &&
+                cand_val.numel() >= 2 * k && cand_idx.numel() >= 2 * k &&
+                out_val.numel() >= k && out_idx.numel() >= k &&
+                hist.numel() >= 2 * HIST_BINS);
+    auto stream = cur_stream();
+    unsigned* h1 = (unsigned*)hist.data_ptr<int>();
+    unsigned* h2 = h1 + HIST_BINS;
+    HIP_OK(hipMemsetAsync(h1, 0, 2 * HIST_BINS * sizeof(int), stream));
+    const long long m = 2 * k;
+    const int blocks = stream_blocks(m);
+    merge_scatter2_kernel<<<blocks, BLOCK, 0, stream>>>(
+        dense.data_ptr<float>(), send_buf.data_ptr<int>(),
+        recv_buf.data_ptr<int>(), (int)k);
+    merge_gather_hist_kernel<<<blocks, BLOCK, 0, stream>>>(
+        dense.data_ptr<float>(), send_buf.data_ptr<int>(),
+        recv_buf.data_ptr<int>(), (int)k, cand_val.data_ptr<float>(),
+        cand_idx.data_ptr<int>(), h1);
+    scan_threshold_kernel<0><<<1, BLOCK, 0, stream>>>(
+        h1, (int)k, info.data_ptr<int>(), counters.data_ptr<int>());
+    refine_hist_kernel<<<blocks, BLOCK, 0, stream>>>(
+        cand_val.data_ptr<float>(), m, info.data_ptr<int>(), h2);
+    scan_threshold_kernel<1><<<1, BLOCK, 0, stream>>>(
+        h2, (int)k, info.data_ptr<int>(), counters.data_ptr<int>());
+    compact_kernel<false, true><<<blocks, BLOCK, 0, stream>>>(
+        cand_val.data_ptr<float>(), cand_idx.data_ptr<int>(), m,
+        info.data_ptr<int>(), counters.data_ptr<int>(),
+        out_val.data_ptr<float>(), out_idx.data_ptr<int>(), (int)k);
+    check_launch("gtopk");
+}
+
 void topk_among(at::Tensor values, at::Tensor indexes, int64_t k,
                 at::Tensor hist, at::Tensor info, at::Tensor counters,
                 at::Tensor out_val, at::Tensor out_idx) {
@@ -562,6 +744,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bn_bwd", &bn_bwd, "fused NHWC BN(+add)(+ReLU) backward");
     m.def("ef_compress", &ef_compress,
           "fused residual accumulate + bucketed top-k select (gfx950)");
+    m.def("ef_hist_zero", &ef_hist_zero,
+          "zero the two-level histogram scratch (overlap path)");
+    m.def("ef_accum_hist_range", &ef_accum_hist_range,
+          "residual[s:e) += grad[s:e) + level-1 histogram (per-bucket, "
+          "launched from backward hooks on the side stream)");
+    m.def("ef_compress_finish", &ef_compress_finish,
+          "threshold scan + refine + compact after per-bucket accumulate");
+    m.def("merge_round", &merge_round,
+          "one fused gtopk merge round: scatter2 + gather/hist + select");
     m.def("topk_among", &topk_among, "top-k among candidate (val,idx) pairs");
     m.def("scatter_add", &scatter_add, "dense[idx] += val");
     m.def("gather_zero", &gather_zero, "out = dense[idx]; dense[idx] = 0");

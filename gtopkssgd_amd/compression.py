@@ -93,6 +93,31 @@ class TopKCompressor:
         self._last[name] = (values, indexes)
         return values, indexes
 
+    def ensure_residual(self, numel: int, device, name: str = "flat"
+                        ) -> torch.Tensor:
+        """Pre-allocate the residual (the backward-overlap path
+        accumulates into it from hooks before compress is ever called)."""
+        residual = self.residuals.get(name)
+        if residual is None or residual.numel() != numel:
+            residual = torch.zeros(numel, dtype=torch.float32, device=device)
+            self.residuals[name] = residual
+        return residual
+
+    def compress_finish(self, name: str = "flat",
+                        ratio: Optional[float] = None,
+                        out_val: Optional[torch.Tensor] = None,
+                        out_idx: Optional[torch.Tensor] = None
+                        ) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Select-only compress: the EF accumulation (and on GPU the
+        level-1 histogram) already ran via ops.ef_accum_hist_range over
+        every range of the flat gradient (backward-overlap path)."""
+        residual = self.residuals[name]
+        k = self.k_for(residual.numel(), ratio)
+        values, indexes = ops.ef_compress_finish(residual, k, out_val,
+                                                 out_idx)
+        self._last[name] = (values, indexes)
+        return values, indexes
+
     def add_residuals(self, included_indexes: torch.Tensor, name: str = "flat"):
         """Re-add locally-selected values whose index is NOT in
         `included_indexes` (the global top-k index set) to the residual."""

@@ -84,6 +84,7 @@ class _Scratch:
         self.info = torch.empty(4, dtype=torch.int32, device=device)
         self.counters = torch.empty(2, dtype=torch.int32, device=device)
         self.marks: Dict[int, torch.Tensor] = {}
+        self.cands: Dict[int, Tuple[torch.Tensor, torch.Tensor]] = {}
 
     def mark_buffer(self, n: int, device: torch.device) -> torch.Tensor:
         buf = self.marks.get(n)
@@ -91,6 +92,15 @@ class _Scratch:
             buf = torch.zeros(n, dtype=torch.uint8, device=device)
             self.marks[n] = buf
         return buf
+
+    def cand_buffers(self, k: int, device: torch.device
+                     ) -> Tuple[torch.Tensor, torch.Tensor]:
+        bufs = self.cands.get(k)
+        if bufs is None:
+            bufs = (torch.empty(2 * k, dtype=torch.float32, device=device),
+                    torch.empty(2 * k, dtype=torch.int32, device=device))
+            self.cands[k] = bufs
+        return bufs
 
 
 _scratch: Dict[int, _Scratch] = {}
@@ -123,6 +133,63 @@ def ef_compress(residual: torch.Tensor, grad: torch.Tensor, k: int
                            out_val, out_idx)
         return out_val, out_idx
     residual.add_(grad)
+    if k == n:
+        values = residual.clone()
+        indexes = torch.arange(n, dtype=torch.int32)
+        residual.zero_()
+        return values, indexes
+    _, idx = torch.topk(residual.abs(), k, sorted=False)
+    values = residual[idx].clone()
+    residual[idx] = 0.0
+    return values, idx.to(torch.int32)
+
+
+# --- backward-overlap variants: the EF accumulate + level-1 histogram
+# is additive over disjoint ranges, so the optimizer launches it
+# per-bucket from post-accumulate-grad hooks on a side HIP stream
+# during backward; ef_compress_finish runs the remaining threshold
+# scan + refine + compact at synchronize() time. ---
+
+def ef_hist_zero(device: torch.device):
+    """Zero the histogram scratch before the first bucket of an
+    iteration (GPU overlap path only)."""
+    s = _get_scratch(device)
+    _ext().ef_hist_zero(s.hist)
+
+
+def ef_accum_hist_range(residual: torch.Tensor, grad: torch.Tensor,
+                        start: int, end: int):
+    """residual[start:end] += grad[start:end] (+ level-1 histogram on
+    GPU).  Disjoint ranges covering [0, n) compose to ef_compress's
+    first pass."""
+    if residual.is_cuda:
+        s = _get_scratch(residual.device)
+        _ext().ef_accum_hist_range(residual, grad, int(start), int(end),
+                                   s.hist)
+        return
+    residual[start:end] += grad[start:end]
+
+
+def ef_compress_finish(residual: torch.Tensor, k: int,
+                       out_val: Optional[torch.Tensor] = None,
+                       out_idx: Optional[torch.Tensor] = None
+                       ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Select after accumulation: (val, idx) = topk_by_magnitude
+    (residual, k); residual[idx] = 0.  On GPU the histogram from the
+    per-range passes must cover the whole vector."""
+    n = residual.numel()
+    assert 0 < k <= n
+    if residual.is_cuda:
+        s = _get_scratch(residual.device)
+        if out_val is None:
+            out_val = torch.empty(k, dtype=torch.float32,
+                                  device=residual.device)
+        if out_idx is None:
+            out_idx = torch.empty(k, dtype=torch.int32,
+                                  device=residual.device)
+        _ext().ef_compress_finish(residual, k, s.hist, s.info, s.counters,
+                                  out_val, out_idx)
+        return out_val, out_idx
     if k == n:
         values = residual.clone()
         indexes = torch.arange(n, dtype=torch.int32)
@@ -179,6 +246,36 @@ def topk_among(values: torch.Tensor, indexes: torch.Tensor, k: int
         return values.clone(), indexes.clone()
     _, pos = torch.topk(values.abs(), k, sorted=False)
     return values[pos].clone(), indexes[pos].clone()
+
+
+def merge_round_(dense: torch.Tensor, send_buf: torch.Tensor,
+                 recv_buf: torch.Tensor, k: int
+                 ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """One fused gtopk merge round (C1 bookkeeping): union-sum both
+    packed sparse sets into ``dense``, gather-and-zero the ≤2k
+    candidates, re-select top-k.  The merged set is written back into
+    ``send_buf`` (wire-ready for the next round); returns (val, idx)
+    views of it.  ``dense`` ends zeroed again.  On GPU this is ONE ext
+    call (7 kernel launches) instead of ~12 — the messages are ~200 KB,
+    latency-bound, so launch count per round is what matters."""
+    sidx = send_buf[:k]
+    sval = send_buf[k:2 * k].view(torch.float32)
+    if dense.is_cuda:
+        s = _get_scratch(dense.device)
+        cand_val, cand_idx = s.cand_buffers(k, dense.device)
+        _ext().merge_round(dense, send_buf, recv_buf, k, s.hist, s.info,
+                           s.counters, cand_val, cand_idx, sval, sidx)
+        return sval, sidx
+    ridx = recv_buf[:k]
+    rval = recv_buf[k:2 * k].view(torch.float32)
+    scatter_add_(dense, sidx, sval)
+    scatter_add_(dense, ridx, rval)
+    cand_idx = torch.cat([sidx, ridx])
+    cand_val = gather_zero(dense, cand_idx)
+    mval, midx = topk_among(cand_val, cand_idx, k)
+    sidx.copy_(midx)
+    sval.copy_(mval)
+    return sval, sidx
 
 
 # ---------------------------------------------------------------------------

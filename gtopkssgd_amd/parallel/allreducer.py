@@ -68,9 +68,6 @@ def gtopk_sparse_allreduce(
         return sval, sidx
     rank_ = dist.get_rank(group)
 
-    ridx, rval = _pack_views(recv_buf, k)
-    # candidate scratch reused across rounds: [0:k)=own idx, [k:2k)=recv
-    cand_idx = torch.empty(2 * k, dtype=torch.int32, device=device)
     group_size = 1
     alive = True
     while group_size < world:
@@ -79,15 +76,9 @@ def gtopk_sparse_allreduce(
                 peer = rank_ + group_size
                 if peer < world:
                     dist.recv(recv_buf, src=peer, group=group)
-                    # merge: union-sum then re-select k of ≤2k candidates
-                    ops.scatter_add_(dense_scratch, sidx, sval)
-                    ops.scatter_add_(dense_scratch, ridx, rval)
-                    cand_idx[:k].copy_(sidx)
-                    cand_idx[k:].copy_(ridx)
-                    cand_val = ops.gather_zero(dense_scratch, cand_idx)
-                    mval, midx = ops.topk_among(cand_val, cand_idx, k)
-                    sidx.copy_(midx)
-                    sval.copy_(mval)
+                    # fused round: union-sum + gather/zero + re-select,
+                    # merged set written back into send_buf (wire-ready)
+                    ops.merge_round_(dense_scratch, send_buf, recv_buf, k)
             else:
                 peer = rank_ - group_size
                 dist.send(send_buf, dst=peer, group=group)

@@ -62,7 +62,8 @@ class _DistributedOptimizer(torch.optim.Optimizer):
                  warmup_iters: int = WARMUP_ITERS,
                  profile: bool = False,
                  writer=None,
-                 overlap_dense: bool = True):
+                 overlap_dense: bool = True,
+                 overlap_sparse: bool = True):
         self.optimizer = optimizer
         self.group = group
         self.err_callback = err_callback
@@ -154,8 +155,34 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         self._dense_handles = []
         self._accum_boundary = True  # False during non-final micro-steps
         self._buckets = []
-        if self.overlap_dense and \
-                (self.mode == "dense" or self.warmup_iters > 0):
+
+        # --- sparse-mode overlap: the EF accumulate + level-1 histogram
+        # is additive per bucket, so it runs on a side HIP stream from
+        # the same hooks during backward; the last bucket also launches
+        # the select finish (threshold scan + refine + compact) there,
+        # leaving only the merge rounds for step() (north star:
+        # compression overlapped with backward on a side stream).
+        self._comm_stream = (torch.cuda.Stream(device=device)
+                             if device.type == "cuda" else None)
+        self._sparse_overlap = (
+            overlap_sparse and device.type == "cuda" and
+            self.mode in ("gtopk", "topk") and
+            hasattr(self.compression, "ensure_residual") and
+            ops.ext_available())
+        self._n_issued = 0
+        self._overlap_select = None
+        self._select_done = None
+        if self._sparse_overlap:
+            self.compression.ensure_residual(self.numel, device)
+            # persistent select outputs (side-stream allocations must not
+            # be recycled by the caching allocator under another stream)
+            self._sel_val = torch.empty(self.k, dtype=torch.float32,
+                                        device=device)
+            self._sel_idx = torch.empty(self.k, dtype=torch.int32,
+                                        device=device)
+        if (self.overlap_dense and
+                (self.mode == "dense" or self.warmup_iters > 0)) or \
+                self._sparse_overlap:
             self._setup_dense_overlap()
 
         self._broadcast_parameters()
@@ -196,21 +223,86 @@ class _DistributedOptimizer(torch.optim.Optimizer):
                 lambda p_, bi=bi: self._on_grad_ready(bi))
 
     def _overlap_active(self) -> bool:
-        return (self._accum_boundary and self._buckets and size() > 1 and
+        return (self.overlap_dense and self._accum_boundary and
+                self._buckets and size() > 1 and
                 (self.mode == "dense" or
                  self.train_iter < self.warmup_iters))
 
+    def _sparse_overlap_active(self) -> bool:
+        return (self._sparse_overlap and self._accum_boundary and
+                self.train_iter >= self.warmup_iters)
+
     def _on_grad_ready(self, bi: int):
-        if not self._overlap_active():
-            return
-        b = self._buckets[bi]
-        b["fired"] += 1
-        if b["fired"] == b["nparams"]:
+        if self._overlap_active():
+            b = self._buckets[bi]
+            b["fired"] += 1
+            if b["fired"] == b["nparams"]:
+                b["fired"] = 0
+                b["issued"] = True
+                view = self.flat_grad[b["start"]:b["end"]]
+                self._dense_handles.append(
+                    dist.all_reduce(view, async_op=True, group=self.group))
+        elif self._sparse_overlap_active():
+            b = self._buckets[bi]
+            b["fired"] += 1
+            if b["fired"] == b["nparams"]:
+                b["fired"] = 0
+                self._issue_sparse_bucket(b)
+
+    # ------------------------------------------------------------------
+    # sparse (gtopk/topk) overlap: per-bucket EF accumulate + histogram
+    # on the side stream during backward; select finish launched by the
+    # last bucket (so at world=1 the whole select precedes step()).
+    def _issue_sparse_bucket(self, b):
+        ev = torch.cuda.Event()
+        ev.record()  # backward stream produced this bucket's grads
+        self._comm_stream.wait_event(ev)
+        with torch.cuda.stream(self._comm_stream):
+            if self._n_issued == 0:
+                ops.ef_hist_zero(self.device)
+            ops.ef_accum_hist_range(
+                self.compression.residuals["flat"], self.flat_grad,
+                b["start"], b["end"])
+        b["issued"] = True
+        self._n_issued += 1
+        if self._n_issued == len(self._buckets):
+            self._launch_select_finish()
+
+    def _launch_select_finish(self):
+        with torch.cuda.stream(self._comm_stream):
+            self._overlap_select = self.compression.compress_finish(
+                "flat", out_val=self._sel_val, out_idx=self._sel_idx)
+        ev = torch.cuda.Event()
+        ev.record(self._comm_stream)
+        self._select_done = ev
+
+    def _finish_sparse_overlap(self):
+        """Returns (vals, idx) if the overlap path ran this iteration,
+        else None (caller falls back to synchronous compress)."""
+        if self._n_issued == 0:
+            return None
+        if self._select_done is None:
+            # some params never fired (unused this iteration):
+            # accumulate their buckets now — backward is done, grads final
+            ev = torch.cuda.Event()
+            ev.record()
+            self._comm_stream.wait_event(ev)
+            with torch.cuda.stream(self._comm_stream):
+                for b in self._buckets:
+                    if not b["issued"]:
+                        ops.ef_accum_hist_range(
+                            self.compression.residuals["flat"],
+                            self.flat_grad, b["start"], b["end"])
+            self._launch_select_finish()
+        torch.cuda.current_stream().wait_event(self._select_done)
+        for b in self._buckets:
+            b["issued"] = False
             b["fired"] = 0
-            b["issued"] = True
-            view = self.flat_grad[b["start"]:b["end"]]
-            self._dense_handles.append(
-                dist.all_reduce(view, async_op=True, group=self.group))
+        self._n_issued = 0
+        result = self._overlap_select
+        self._overlap_select = None
+        self._select_done = None
+        return result
 
     def _finish_dense_overlap(self) -> bool:
         """Wait outstanding bucket all-reduces; returns True when the
@@ -292,6 +384,12 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         """Whole-model grad clipping as ONE norm over the flat buffer
         (equivalent to torch.nn.utils.clip_grad_norm_ over all params,
         without per-tensor kernel launches)."""
+        if self._n_issued > 0:
+            raise RuntimeError(
+                "clip_grad_norm_ after the backward-overlap path already "
+                "folded unscaled gradients into the EF residual; construct "
+                "DistributedOptimizer with overlap_sparse=False when "
+                "gradient clipping is used")
         total = torch.linalg.vector_norm(self.flat_grad)
         scale = torch.clamp(max_norm / (total + 1e-6), max=1.0)
         self.flat_grad.mul_(scale)  # no host sync: clamp keeps <=1
@@ -322,7 +420,12 @@ class _DistributedOptimizer(torch.optim.Optimizer):
             self._staged = ("dense", None, None)
         elif self.mode == "gtopk":
             pt.start("compress")
-            vals, idx = self.compression.compress(self.flat_grad, "flat")
+            sel = self._finish_sparse_overlap() if self._sparse_overlap \
+                else None
+            if sel is None:
+                vals, idx = self.compression.compress(self.flat_grad, "flat")
+            else:
+                vals, idx = sel
             pt.stop("compress")
             pt.start("comm")
             gval, gidx = self.reducer.gtopk(vals, idx)
@@ -333,7 +436,12 @@ class _DistributedOptimizer(torch.optim.Optimizer):
             self._staged = ("sparse", gval, gidx)
         else:  # topk baseline: union sum sits in dense_scratch
             pt.start("compress")
-            vals, idx = self.compression.compress(self.flat_grad, "flat")
+            sel = self._finish_sparse_overlap() if self._sparse_overlap \
+                else None
+            if sel is None:
+                vals, idx = self.compression.compress(self.flat_grad, "flat")
+            else:
+                vals, idx = sel
             pt.stop("compress")
             pt.start("comm")
             self.reducer.topk(vals, idx)
@@ -422,11 +530,12 @@ def DistributedOptimizer(optimizer, named_parameters=None, compression=None,
                          err_callback=None,
                          warmup_iters: int = WARMUP_ITERS,
                          profile: bool = False, writer=None,
-                         overlap_dense: bool = True):
+                         overlap_dense: bool = True,
+                         overlap_sparse: bool = True):
     """Factory mirroring the reference's
     ``DistributedOptimizer(optimizer, named_parameters, compression,
     is_sparse, density, err_callback, ...)``."""
     return _DistributedOptimizer(optimizer, named_parameters, compression,
                                  is_sparse, density, mode, group,
                                  err_callback, warmup_iters, profile, writer,
-                                 overlap_dense)
+                                 overlap_dense, overlap_sparse)

@@ -109,7 +109,10 @@ def main():
     opt = gt.DistributedOptimizer(
         trainer.optimizer, trainer.net.named_parameters(),
         compression=compressor, density=args.density,
-        profile=args.profile or bool(writer), writer=writer)
+        profile=args.profile or bool(writer), writer=writer,
+        # clipping rescales grads after backward — incompatible with the
+        # backward-overlapped EF accumulate
+        overlap_sparse=args.norm_clip is None)
 
     start_epoch = 0
     if args.resume:
