@@ -212,6 +212,10 @@ def _self_benchmark():  # pragma: no cover - exercised on GPU boxes
     p.add_argument("--mode", default="gtopk",
                    choices=["gtopk", "topk", "dense", "torchtopk"])
     p.add_argument("--device", default=None)
+    p.add_argument("--simulate-rounds", type=int, default=0,
+                   help="world=1: run R fused merge-round bookkeeping "
+                        "passes per iteration (per-round kernel cost "
+                        "without comm — the xGMI p2p time adds on top)")
     args = p.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -234,6 +238,16 @@ def _self_benchmark():  # pragma: no cover - exercised on GPU boxes
     g = torch.Generator(device="cpu").manual_seed(rank)
     grad = torch.randn(n, generator=g).to(device)
 
+    if args.simulate_rounds:
+        # a synthetic peer message: a valid packed sparse set from a
+        # different random vector
+        gp = torch.Generator(device="cpu").manual_seed(rank + 777)
+        peer = torch.randn(n, generator=gp)
+        pidx = torch.topk(peer.abs(), k, sorted=False).indices
+        reducer.recv_buf[:k].copy_(pidx.to(torch.int32).to(device))
+        reducer.recv_buf[k:].view(torch.float32).copy_(
+            peer[pidx].to(device))
+
     def one_iter():
         if args.mode == "dense":
             reducer.dense(grad.clone())
@@ -248,6 +262,15 @@ def _self_benchmark():  # pragma: no cover - exercised on GPU boxes
             return
         vals, idx = comp.compress(grad, "b")
         if args.mode == "gtopk":
+            if args.simulate_rounds and world == 1:
+                sb = reducer.send_buf
+                sb[:k].copy_(idx)
+                sb[k:].view(torch.float32).copy_(vals)
+                for _ in range(args.simulate_rounds):
+                    ops.merge_round_(reducer.dense_scratch, sb,
+                                     reducer.recv_buf, k)
+                comp.add_residuals(sb[:k], "b")
+                return
             gval, gidx = reducer.gtopk(vals, idx)
             comp.add_residuals(gidx, "b")
         else:
@@ -272,6 +295,7 @@ def _self_benchmark():  # pragma: no cover - exercised on GPU boxes
         print(json.dumps({
             "collective": args.mode, "n": n, "k": k, "world": world,
             "ms_per_iter": round(dt * 1000, 3),
+            "simulate_rounds": args.simulate_rounds,
             "device": str(device)}), flush=True)
     if world > 1:
         dist.destroy_process_group()
