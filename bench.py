@@ -133,8 +133,12 @@ def main():
                             weight_decay=1e-4)
     comp = (gt.compressors["none"]() if args.mode == "dense"
             else gt.compressors[args.mode](args.density))
-    opt = gt.DistributedOptimizer(inner, model.named_parameters(),
-                                  compression=comp, density=args.density)
+    opt = gt.DistributedOptimizer(
+        inner, model.named_parameters(), compression=comp,
+        density=args.density,
+        # GTOPK_OVERLAP=0 disables the backward-overlapped compression
+        # (A/B knob; default on)
+        overlap_sparse=os.environ.get("GTOPK_OVERLAP", "1") == "1")
     crit = nn.CrossEntropyLoss()
 
     n_buf = 4  # pre-staged synthetic batches, cycled

@@ -184,6 +184,8 @@ class _DistributedOptimizer(torch.optim.Optimizer):
                 (self.mode == "dense" or self.warmup_iters > 0)) or \
                 self._sparse_overlap:
             self._setup_dense_overlap()
+        self._hook_mode = 0
+        self._refresh_hook_mode()
 
         self._broadcast_parameters()
         logger.info(
@@ -232,22 +234,33 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         return (self._sparse_overlap and self._accum_boundary and
                 self.train_iter >= self.warmup_iters)
 
-    def _on_grad_ready(self, bi: int):
+    def _refresh_hook_mode(self):
+        """Precompute the per-iteration hook dispatch (0 off / 1 dense /
+        2 sparse) so the per-parameter hook — called len(params) times
+        inside every backward — is one attribute load + int test."""
         if self._overlap_active():
-            b = self._buckets[bi]
-            b["fired"] += 1
-            if b["fired"] == b["nparams"]:
-                b["fired"] = 0
-                b["issued"] = True
-                view = self.flat_grad[b["start"]:b["end"]]
-                self._dense_handles.append(
-                    dist.all_reduce(view, async_op=True, group=self.group))
+            self._hook_mode = 1
         elif self._sparse_overlap_active():
-            b = self._buckets[bi]
-            b["fired"] += 1
-            if b["fired"] == b["nparams"]:
-                b["fired"] = 0
-                self._issue_sparse_bucket(b)
+            self._hook_mode = 2
+        else:
+            self._hook_mode = 0
+
+    def _on_grad_ready(self, bi: int):
+        mode = self._hook_mode
+        if not mode:
+            return
+        b = self._buckets[bi]
+        b["fired"] += 1
+        if b["fired"] < b["nparams"]:
+            return
+        b["fired"] = 0
+        if mode == 1:
+            b["issued"] = True
+            view = self.flat_grad[b["start"]:b["end"]]
+            self._dense_handles.append(
+                dist.all_reduce(view, async_op=True, group=self.group))
+        else:
+            self._issue_sparse_bucket(b)
 
     # ------------------------------------------------------------------
     # sparse (gtopk/topk) overlap: per-bucket EF accumulate + histogram
@@ -330,6 +343,7 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         before non-final micro-step backwards so dense-overlap hooks do
         not all-reduce partial gradients; True before the final one."""
         self._accum_boundary = bool(is_boundary)
+        self._refresh_hook_mode()
 
     @staticmethod
     def _can_fuse(optimizer: torch.optim.Optimizer) -> bool:
@@ -500,6 +514,7 @@ class _DistributedOptimizer(torch.optim.Optimizer):
                 self.writer.add_scalar(f"opt/{tag}_ms", ms, self.train_iter)
         self.train_iter += 1
         self._synchronized = False
+        self._refresh_hook_mode()  # warmup->sparse transition etc.
         return loss
 
     # ------------------------------------------------------------------
