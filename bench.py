@@ -32,6 +32,7 @@ _CONFIGS = {
     "resnet20": dict(shape=(3, 32, 32), classes=10, batch=2048),
     "vgg16": dict(shape=(3, 32, 32), classes=10, batch=1024),
     "lstm": dict(seq_len=35, vocab=10000, batch=512),
+    "lstman4": dict(freq=161, time=200, labels=29, tmax=20, batch=64),
 }
 
 
@@ -122,7 +123,8 @@ def main():
     else:
         torch.backends.cudnn.benchmark = True
     is_lm = args.model == "lstm"
-    channels_last = not is_lm and not args.no_channels_last
+    is_ctc = args.model == "lstman4"
+    channels_last = not (is_lm or is_ctc) and not args.no_channels_last
     if is_lm:
         model = zoo.lstm_ptb(cfg["vocab"], 1500).to(device)
     else:
@@ -139,7 +141,8 @@ def main():
         # GTOPK_OVERLAP=0 disables the backward-overlapped compression
         # (A/B knob; default on)
         overlap_sparse=os.environ.get("GTOPK_OVERLAP", "1") == "1")
-    crit = nn.CrossEntropyLoss()
+    crit = nn.CTCLoss(blank=0, zero_infinity=True) if is_ctc \
+        else nn.CrossEntropyLoss()
 
     n_buf = 4  # pre-staged synthetic batches, cycled
     gen = torch.Generator(device="cpu").manual_seed(4321 + rank)
@@ -147,6 +150,13 @@ def main():
         xs = [torch.randint(0, cfg["vocab"], (cfg["seq_len"] + 1, B),
                             generator=gen).to(device) for _ in range(n_buf)]
         hidden0 = model.init_hidden(B)
+    elif is_ctc:
+        xs = [torch.randn(B, 1, cfg["freq"], cfg["time"], generator=gen)
+              .to(device) for _ in range(n_buf)]
+        tlen = cfg["tmax"]
+        ys = [torch.randint(1, cfg["labels"], (B * tlen,), generator=gen)
+              .to(device) for _ in range(n_buf)]
+        target_lengths = torch.full((B,), tlen, dtype=torch.long)
     else:
         xs = [torch.randn(B, *cfg["shape"], generator=gen)
               .to(device, memory_format=(torch.channels_last if channels_last
@@ -163,6 +173,13 @@ def main():
                 logits, _ = model(toks[:-1], hidden0)
                 loss = crit(logits.reshape(-1, logits.size(-1)),
                             toks[1:].reshape(-1))
+            elif is_ctc:
+                logits = model(xs[i % n_buf])  # [T, B, L]
+                log_probs = logits.log_softmax(2).float()
+                t = logits.size(0)
+                input_lengths = torch.full((B,), t, dtype=torch.long)
+                loss = crit(log_probs, ys[i % n_buf].view(B, -1),
+                            input_lengths, target_lengths)
             else:
                 loss = crit(model(xs[i % n_buf]), ys[i % n_buf])
         loss.backward()
@@ -197,7 +214,8 @@ def main():
     if rank == 0:
         per_step_items = B * (cfg["seq_len"] if is_lm else 1)
         value = world * per_step_items * args.steps / elapsed
-        unit = "tokens/sec" if is_lm else "images/sec"
+        unit = ("tokens/sec" if is_lm else
+                "utterances/sec" if is_ctc else "images/sec")
         metric = (f"{unit} (whole node) {args.model} {args.mode} "
                   f"density={args.density}")
         if args.model == "resnet50" and args.mode == "gtopk" \

@@ -77,27 +77,47 @@ class SyntheticPTB(Dataset):
 
 
 class SyntheticAN4(Dataset):
-    """Random spectrograms + random transcripts for the CTC path."""
+    """Random spectrograms + random transcripts for the CTC path.
+
+    ``learnable=True`` renders each transcript label as a distinctive
+    per-label frequency pattern over consecutive time segments (SNR ~2),
+    so CTC loss / WER / CER curves are meaningful — the AN4 analogue of
+    SyntheticImages' class-mean task."""
 
     def __init__(self, freq_bins: int = 161, time_steps: int = 200,
                  max_transcript: int = 20, n_labels: int = 29,
-                 n_samples: int = 948, seed: int = 0):
+                 n_samples: int = 948, seed: int = 0,
+                 learnable: bool = False):
         self.freq_bins = freq_bins
         self.time_steps = time_steps
         self.max_transcript = max_transcript
         self.n_labels = n_labels
         self.n = n_samples
         self.seed = seed
+        self.learnable = learnable
+        if learnable:
+            g = torch.Generator().manual_seed(seed * 31 + 7)
+            self.label_patterns = torch.randn(n_labels, freq_bins,
+                                              generator=g)
 
     def __len__(self):
         return self.n
 
     def __getitem__(self, i):
         g = torch.Generator().manual_seed(self.seed * 13_000_003 + i)
-        x = torch.randn(1, self.freq_bins, self.time_steps, generator=g)
-        tlen = int(torch.randint(5, self.max_transcript, (1,), generator=g))
+        lo = max(1, min(5, self.max_transcript - 1))
+        tlen = int(torch.randint(lo, self.max_transcript, (1,), generator=g))
         # labels 1..n_labels-1 (0 is the CTC blank)
         y = torch.randint(1, self.n_labels, (tlen,), generator=g)
+        if not self.learnable:
+            x = torch.randn(1, self.freq_bins, self.time_steps, generator=g)
+            return x, y
+        x = torch.randn(1, self.freq_bins, self.time_steps,
+                        generator=g) * 0.5
+        seg = self.time_steps // tlen
+        for j, lab in enumerate(y.tolist()):
+            x[0, :, j * seg:(j + 1) * seg] += \
+                self.label_patterns[lab].unsqueeze(1)
         return x, y
 
 

@@ -245,14 +245,35 @@ class DLTrainer:
             logger.info("[epoch %d] test perplexity %.2f", epoch, ppl)
             return self._sync_metrics({"perplexity": ppl})
         if self.is_ctc:
+            from .models import GreedyDecoder
+            decoder = GreedyDecoder()
             total_loss, count = 0.0, 0
-            for batch in self.testloader:
-                loss = self._forward_loss(batch)
-                total_loss += loss.item()
+            wer_sum, cer_sum, n_utt = 0.0, 0.0, 0
+            for x, ys in self.testloader:
+                x = x.to(self.device)
+                logits = self.net(x)  # [T, B, L]
+                log_probs = logits.log_softmax(2)
+                t, b, _ = log_probs.shape
+                input_lengths = torch.full((b,), t, dtype=torch.long)
+                target_lengths = torch.tensor([len(y) for y in ys],
+                                              dtype=torch.long)
+                targets = torch.cat([torch.as_tensor(y) for y in ys])
+                total_loss += self.criterion(
+                    log_probs, targets, input_lengths,
+                    target_lengths).item()
                 count += 1
+                for hyp, y in zip(decoder.decode(logits), ys):
+                    ref = decoder.labels_to_string(y)
+                    wer_sum += decoder.wer(hyp, ref)
+                    cer_sum += decoder.cer(hyp, ref)
+                    n_utt += 1
             avg = total_loss / max(1, count)
-            logger.info("[epoch %d] test ctc-loss %.4f", epoch, avg)
-            return self._sync_metrics({"ctc_loss": avg})
+            wer = wer_sum / max(1, n_utt)
+            cer = cer_sum / max(1, n_utt)
+            logger.info("[epoch %d] test ctc-loss %.4f wer %.4f cer %.4f",
+                        epoch, avg, wer, cer)
+            return self._sync_metrics({"ctc_loss": avg, "wer": wer,
+                                       "cer": cer})
         top1, top5, total = 0, 0, 0
         for x, y in self.testloader:
             x = x.to(self.device)
