@@ -108,16 +108,17 @@ def main():
     B = args.batch_size or cfg["batch"]
     # MIOpen solver selection.  Measured on MI355X (profiles/RESULTS.md):
     # benchmark=True re-runs the ~115 s exhaustive find in EVERY process
-    # (the find-db hit does not skip it), while FAST find mode with the
-    # committed tuned db gives identical steady-state throughput (8585.2
-    # vs 8585.9 img/s ResNet-50 bs512) with ~3 s warmup.  So: FAST when
-    # the db snapshot covers this exact config (the default batch sizes
-    # it was captured with), exhaustive find otherwise (FAST without a
-    # tuned db falls back to naive solvers — 209 img/s).
+    # (neither the user find-db nor a same-box second process skips it),
+    # while FAST find mode gives ResNet-50 throughput parity on a FRESH
+    # box (8675.7 img/s, 0.7 s warmup vs 8656.0, 115 s) — its ImageNet
+    # shapes are covered by MIOpen's system databases.  The CIFAR-shape
+    # models and AlexNet are NOT covered (FAST fresh-box: resnet20 1850
+    # vs 276547 img/s), so they keep the exhaustive find (their finds
+    # are 36-50 s).  GTOPK_CUDNN_BENCHMARK=0/1 forces either path.
     force = os.environ.get("GTOPK_CUDNN_BENCHMARK")
     if force is not None:
         torch.backends.cudnn.benchmark = force == "1"
-    elif primed and B == cfg["batch"]:
+    elif args.model == "resnet50" and B == cfg["batch"]:
         os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
         torch.backends.cudnn.benchmark = False
     else:
