@@ -6,12 +6,17 @@ ResNet-50 bs512.  Candidate costs: (a) per-param Python hook calls,
 (b) per-bucket event create/record/wait + stream switches, (c) the
 side-stream kernels contending with backward for HBM/L2.
 
+Measured verdict (2026-09-13): (b) — the side stream + events — was the
+whole 1.2 ms; hooks and kernels were free.  The shipped implementation
+now launches the bucket kernels on the current (backward) stream, which
+matched overlap-off throughput; the "sidestream" variant here
+reproduces the abandoned design for the record.
+
 Variants:
-  off       overlap disabled (sync select in step)
-  noop      hooks fire but buckets do nothing (isolates (a))
-  defstream hooks launch EF+hist+finish on the DEFAULT stream
-            (isolates (b): no events, no side stream; keeps (c)-ish)
-  full      the shipped side-stream overlap
+  off        overlap disabled (sync select in step)
+  noop       hooks fire but buckets do nothing (isolates (a))
+  full       the shipped in-stream hook overlap
+  sidestream the abandoned side-stream + events design
 """
 import json
 import os
@@ -51,21 +56,37 @@ def run(variant, steps=30, warmup=10, batch=512):
         overlap_sparse=variant != "off")
     if variant == "noop":
         opt._issue_sparse_bucket = lambda b: None
-    elif variant == "defstream":
+    elif variant == "sidestream":
+        stream = torch.cuda.Stream()
+        state = {"ev": None}
+
         def issue(b):
-            if opt._n_issued == 0:
-                ops.ef_hist_zero(opt.device)
-            ops.ef_accum_hist_range(opt.compression.residuals["flat"],
-                                    opt.flat_grad, b["start"], b["end"])
+            ev = torch.cuda.Event()
+            ev.record()
+            stream.wait_event(ev)
+            with torch.cuda.stream(stream):
+                if opt._n_issued == 0:
+                    ops.ef_hist_zero(opt.device)
+                ops.ef_accum_hist_range(opt.compression.residuals["flat"],
+                                        opt.flat_grad, b["start"], b["end"])
             b["issued"] = True
             opt._n_issued += 1
             if opt._n_issued == len(opt._buckets):
-                opt._overlap_select = opt.compression.compress_finish(
-                    "flat", out_val=opt._sel_val, out_idx=opt._sel_idx)
-                ev = torch.cuda.Event()
-                ev.record()
-                opt._select_done = ev
+                with torch.cuda.stream(stream):
+                    opt._overlap_select = opt.compression.compress_finish(
+                        "flat", out_val=opt._sel_val, out_idx=opt._sel_idx)
+                done = torch.cuda.Event()
+                done.record(stream)
+                state["ev"] = done
         opt._issue_sparse_bucket = issue
+        orig_finish = opt._finish_sparse_overlap
+
+        def finish():
+            if state["ev"] is not None:
+                torch.cuda.current_stream().wait_event(state["ev"])
+                state["ev"] = None
+            return orig_finish()
+        opt._finish_sparse_overlap = finish
     crit = nn.CrossEntropyLoss()
     gen = torch.Generator().manual_seed(1)
     xs = [torch.randn(batch, 3, 224, 224, generator=gen)
@@ -98,6 +119,6 @@ def run(variant, steps=30, warmup=10, batch=512):
 
 if __name__ == "__main__":
     prime_db()
-    order = ["off", "full", "noop", "defstream", "off", "full"]
+    order = ["off", "full", "noop", "sidestream", "off", "full"]
     for v in order:
         run(v)

@@ -157,13 +157,16 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         self._buckets = []
 
         # --- sparse-mode overlap: the EF accumulate + level-1 histogram
-        # is additive per bucket, so it runs on a side HIP stream from
-        # the same hooks during backward; the last bucket also launches
-        # the select finish (threshold scan + refine + compact) there,
-        # leaving only the merge rounds for step() (north star:
-        # compression overlapped with backward on a side stream).
-        self._comm_stream = (torch.cuda.Stream(device=device)
-                             if device.type == "cuda" else None)
+        # is additive per bucket, so it is launched per 25 MB bucket from
+        # the post-accumulate-grad hooks as backward produces each
+        # bucket; the last bucket also launches the select finish
+        # (threshold scan + refine + compact), leaving only the merge
+        # rounds for step().  The kernels go on the CURRENT (backward)
+        # stream: stream order already guarantees the bucket's grads are
+        # final, and a dedicated side stream + events was measured at
+        # +1.2 ms/step of launch-pipeline interference with zero GPU-time
+        # win (profiles/RESULTS.md, benchmarks/probe_overlap.py ablation:
+        # off 58.5 / side-stream 59.7 / in-stream hooks 58.5 ms/step).
         self._sparse_overlap = (
             overlap_sparse and device.type == "cuda" and
             self.mode in ("gtopk", "topk") and
@@ -171,7 +174,6 @@ class _DistributedOptimizer(torch.optim.Optimizer):
             ops.ext_available())
         self._n_issued = 0
         self._overlap_select = None
-        self._select_done = None
         if self._sparse_overlap:
             self.compression.ensure_residual(self.numel, device)
             # persistent select outputs (side-stream allocations must not
@@ -264,57 +266,44 @@ class _DistributedOptimizer(torch.optim.Optimizer):
 
     # ------------------------------------------------------------------
     # sparse (gtopk/topk) overlap: per-bucket EF accumulate + histogram
-    # on the side stream during backward; select finish launched by the
-    # last bucket (so at world=1 the whole select precedes step()).
+    # launched from the hooks mid-backward (current stream); select
+    # finish launched by the last bucket, so the whole local select is
+    # enqueued — and typically executed — before step() is entered.
     def _issue_sparse_bucket(self, b):
-        ev = torch.cuda.Event()
-        ev.record()  # backward stream produced this bucket's grads
-        self._comm_stream.wait_event(ev)
-        with torch.cuda.stream(self._comm_stream):
-            if self._n_issued == 0:
-                ops.ef_hist_zero(self.device)
-            ops.ef_accum_hist_range(
-                self.compression.residuals["flat"], self.flat_grad,
-                b["start"], b["end"])
+        if self._n_issued == 0:
+            ops.ef_hist_zero(self.device)
+        ops.ef_accum_hist_range(
+            self.compression.residuals["flat"], self.flat_grad,
+            b["start"], b["end"])
         b["issued"] = True
         self._n_issued += 1
         if self._n_issued == len(self._buckets):
             self._launch_select_finish()
 
     def _launch_select_finish(self):
-        with torch.cuda.stream(self._comm_stream):
-            self._overlap_select = self.compression.compress_finish(
-                "flat", out_val=self._sel_val, out_idx=self._sel_idx)
-        ev = torch.cuda.Event()
-        ev.record(self._comm_stream)
-        self._select_done = ev
+        self._overlap_select = self.compression.compress_finish(
+            "flat", out_val=self._sel_val, out_idx=self._sel_idx)
 
     def _finish_sparse_overlap(self):
         """Returns (vals, idx) if the overlap path ran this iteration,
         else None (caller falls back to synchronous compress)."""
         if self._n_issued == 0:
             return None
-        if self._select_done is None:
+        if self._overlap_select is None:
             # some params never fired (unused this iteration):
             # accumulate their buckets now — backward is done, grads final
-            ev = torch.cuda.Event()
-            ev.record()
-            self._comm_stream.wait_event(ev)
-            with torch.cuda.stream(self._comm_stream):
-                for b in self._buckets:
-                    if not b["issued"]:
-                        ops.ef_accum_hist_range(
-                            self.compression.residuals["flat"],
-                            self.flat_grad, b["start"], b["end"])
+            for b in self._buckets:
+                if not b["issued"]:
+                    ops.ef_accum_hist_range(
+                        self.compression.residuals["flat"],
+                        self.flat_grad, b["start"], b["end"])
             self._launch_select_finish()
-        torch.cuda.current_stream().wait_event(self._select_done)
         for b in self._buckets:
             b["issued"] = False
             b["fired"] = 0
         self._n_issued = 0
         result = self._overlap_select
         self._overlap_select = None
-        self._select_done = None
         return result
 
     def _finish_dense_overlap(self) -> bool:
