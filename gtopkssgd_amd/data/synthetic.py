@@ -131,6 +131,12 @@ def make_dataset(dataset: str, n_samples: int = None, seed: int = 0,
                             **kwargs)
     if dataset == "an4":
         return SyntheticAN4(n_samples=n_samples or 948, seed=seed, **kwargs)
+    if dataset == "an4wav":
+        # full audio front-end: tone-sequence waveforms through the real
+        # STFT log-spectrogram pipeline (data/audio.py)
+        from .audio import SyntheticAN4Waveforms
+        return SyntheticAN4Waveforms(n_samples=n_samples or 948, seed=seed,
+                                     **kwargs)
     raise ValueError(f"unknown dataset '{dataset}'")
 
 
@@ -150,7 +156,14 @@ def make_loader(dataset, batch_size: int, rank: int = 0, world: int = 1,
         sampler = DistributedSampler(dataset, num_replicas=world, rank=rank,
                                      shuffle=True, seed=seed,
                                      drop_last=drop_last)
-    collate = _ctc_collate if isinstance(dataset, SyntheticAN4) else None
+    from .audio import SpectrogramDataset, SyntheticAN4Waveforms, \
+        ctc_collate_padded
+    if isinstance(dataset, (SpectrogramDataset, SyntheticAN4Waveforms)):
+        collate = ctc_collate_padded  # variable-length time axis
+    elif isinstance(dataset, SyntheticAN4):
+        collate = _ctc_collate
+    else:
+        collate = None
     return DataLoader(dataset, batch_size=batch_size, sampler=sampler,
                       shuffle=(sampler is None), drop_last=drop_last,
                       num_workers=num_workers, pin_memory=True,

@@ -33,6 +33,7 @@ _LR_MILESTONES = {
     "mnist": [],
     "ptb": [20, 30],
     "an4": [],
+    "an4wav": [],
 }
 
 
