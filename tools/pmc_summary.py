@@ -1,54 +1,47 @@
 #!/usr/bin/env python3
-"""Summarise a rocprofv3 --pmc FETCH_SIZE WRITE_SIZE run into per-kernel
-effective HBM traffic + bandwidth (needs the matching kernel durations
-from a --stats run or the counter file's own timestamps).
+"""Summarise rocprofv3 --pmc results (rocpd SQLite .db output) into a
+per-kernel DRAM-traffic table: dispatches, counter GB, Σduration,
+effective rate.
 
-Usage: python tools/pmc_summary.py gpurun_out/pmc > profiles/pmc_summary.md
+rocprofv3 7.2 on gfx950 writes `<out>_results.db`; counters must be
+collected ONE per run (FETCH_SIZE + WRITE_SIZE in one pass exceeds the
+hardware counter capacity).  FETCH_SIZE/WRITE_SIZE are in KB
+(documented expression: TCC_EA0 request counts × line size / 1024).
+
+Usage: python tools/pmc_summary.py gpurun_out/pmc/sel_FETCH_SIZE_results.db
 """
-import csv
-import glob
+import re
+import sqlite3
 import sys
-from collections import defaultdict
 
 
-def main(d):
-    files = glob.glob(f"{d}/**/*counter_collection.csv", recursive=True) \
-        + glob.glob(f"{d}/*counter_collection.csv")
-    if not files:
-        print(f"no counter_collection.csv under {d}", file=sys.stderr)
-        return 1
-    # per-kernel: accumulate counter values and dispatch durations
-    acc = defaultdict(lambda: defaultdict(float))
-    dur = defaultdict(float)
-    count = defaultdict(set)
-    for f in files:
-        with open(f) as fh:
-            for row in csv.DictReader(fh):
-                name = row.get("Kernel_Name", row.get("Kernel-Name", ""))
-                short = name.split("(")[0].split("<")[0].strip()
-                cname = row.get("Counter_Name", row.get("Counter-Name"))
-                val = float(row.get("Counter_Value",
-                                    row.get("Counter-Value", 0)))
-                acc[short][cname] += val
-                did = row.get("Dispatch_Id", row.get("Dispatch-Id"))
-                if did is not None and did not in count[short]:
-                    count[short].add(did)
-                    s = row.get("Start_Timestamp")
-                    e = row.get("End_Timestamp")
-                    if s and e:
-                        dur[short] += (float(e) - float(s)) / 1e3  # ns->us
-    print("| kernel | dispatches | fetch GB | write GB | Σus | TB/s |")
-    print("|---|---|---|---|---|---|")
-    for k in sorted(acc, key=lambda k: -dur[k]):
-        # FETCH_SIZE/WRITE_SIZE are reported in KB per dispatch
-        fetch = acc[k].get("FETCH_SIZE", 0) / 1024 / 1024  # KB -> GB
-        write = acc[k].get("WRITE_SIZE", 0) / 1024 / 1024
-        us = dur[k]
-        bw = (fetch + write) / (us / 1e6) / 1024 if us else 0  # GB/s->TB/s
-        print(f"| {k[:60]} | {len(count[k])} | {fetch:.2f} | {write:.2f} "
-              f"| {us:.0f} | {bw:.2f} |")
-    return 0
+def summarize(db):
+    c = sqlite3.connect(db)
+    names = [r[0] for r in c.execute(
+        "SELECT name FROM sqlite_master WHERE name LIKE 'rocpd_pmc_event%'")]
+    if not names:
+        raise SystemExit(f"{db}: no rocpd_pmc_event table")
+    sfx = names[0].replace('rocpd_pmc_event_', '')
+    counter = c.execute(
+        f"SELECT name FROM rocpd_info_pmc_{sfx}").fetchone()[0]
+    q = f"""
+    SELECT ks.display_name, count(*), sum(pe.value),
+           sum(kd.end-kd.start)/1e3
+    FROM rocpd_pmc_event_{sfx} pe
+    JOIN rocpd_kernel_dispatch_{sfx} kd ON kd.event_id = pe.event_id
+    JOIN rocpd_info_kernel_symbol_{sfx} ks ON ks.id = kd.kernel_id
+    GROUP BY ks.display_name ORDER BY 4 DESC
+    """
+    print(f"### {db} — counter {counter} (KB units)")
+    print(f"| kernel | dispatches | {counter} GB | Σus | TB/s |")
+    print("|---|---|---|---|---|")
+    for name, n, val, us in c.execute(q):
+        short = re.split(r'[<(]', name)[0].strip()[:48]
+        gb = val / 1024 / 1024
+        bw = gb / (us / 1e6) / 1024 if us else 0
+        print(f"| {short} | {n} | {gb:.2f} | {us:.0f} | {bw:.2f} |")
 
 
 if __name__ == "__main__":
-    sys.exit(main(sys.argv[1] if len(sys.argv) > 1 else "gpurun_out/pmc"))
+    for db in sys.argv[1:]:
+        summarize(db)
