@@ -27,7 +27,9 @@ from gtopkssgd_amd import models as zoo  # noqa: E402
 
 # model -> (builder kwargs dataset-shape, default per-GPU batch, unit)
 _CONFIGS = {
-    "resnet50": dict(shape=(3, 224, 224), classes=1000, batch=512),
+    # bs768 measured 9200 img/s vs bs512's 8861 under the same FAST
+    # find mode + committed MIOpen db (0.8 s warmup) — RESULTS.md sweep
+    "resnet50": dict(shape=(3, 224, 224), classes=1000, batch=768),
     "alexnet": dict(shape=(3, 224, 224), classes=1000, batch=512),
     "resnet20": dict(shape=(3, 32, 32), classes=10, batch=2048),
     "vgg16": dict(shape=(3, 32, 32), classes=10, batch=1024),
@@ -118,7 +120,7 @@ def main():
     force = os.environ.get("GTOPK_CUDNN_BENCHMARK")
     if force is not None:
         torch.backends.cudnn.benchmark = force == "1"
-    elif args.model == "resnet50" and B == cfg["batch"]:
+    elif args.model == "resnet50" and B in (256, 512, 768):
         os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
         torch.backends.cudnn.benchmark = False
     else:
