@@ -76,17 +76,24 @@ def _prime_miopen_db() -> bool:
 
 def main():
     args = parse_args()
-    assert torch.cuda.is_available(), "bench.py requires an MI355X GPU"
+    # GTOPK_BENCH_CPU=1: run the identical code path on CPU over gloo —
+    # a plumbing test of the multi-rank bench (env parsing, barriers,
+    # MAX-reduce, JSON contract), NOT a performance mode.  The driver's
+    # GPU runs never set it.
+    cpu_mode = os.environ.get("GTOPK_BENCH_CPU") == "1"
+    if not cpu_mode:
+        assert torch.cuda.is_available(), "bench.py requires an MI355X GPU"
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     if world > 1:
-        dist.init_process_group(backend="nccl")
-        torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend="gloo" if cpu_mode else "nccl")
+        if not cpu_mode:
+            torch.cuda.set_device(local_rank)
     # rank 0 extracts the MIOpen cache snapshot; others wait (a
     # concurrent multi-rank extract would race on the same files)
     primed = False
-    if local_rank == 0:
+    if local_rank == 0 and not cpu_mode:
         primed = _prime_miopen_db()
     if world > 1:
         dist.barrier()
@@ -103,7 +110,8 @@ def main():
             except OSError as exc:
                 print(f"# per-rank miopen db copy failed: {exc}",
                       flush=True)
-    device = torch.device("cuda", local_rank)
+    device = torch.device("cpu") if cpu_mode \
+        else torch.device("cuda", local_rank)
     torch.manual_seed(1234)
 
     cfg = _CONFIGS[args.model]
@@ -170,7 +178,8 @@ def main():
 
     def step(i):
         opt.zero_grad()
-        with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+        with torch.autocast(device_type=device.type,
+                            dtype=torch.bfloat16, enabled=not cpu_mode):
             if is_lm:
                 toks = xs[i % n_buf]
                 logits, _ = model(toks[:-1], hidden0)
@@ -189,23 +198,27 @@ def main():
         opt.step()
         return loss
 
+    def _sync():
+        if not cpu_mode:
+            torch.cuda.synchronize()
+
     model.train()
     tw = time.perf_counter()
     for i in range(args.warmup):
         step(i)
-    torch.cuda.synchronize()
+    _sync()
     if rank == 0:
         print(f"# warmup: {time.perf_counter() - tw:.1f}s "
               f"({args.warmup} steps)", flush=True)
     if world > 1:
         dist.barrier()
-    torch.cuda.synchronize()
+    _sync()
     t0 = time.perf_counter()
     for i in range(args.steps):
         step(args.warmup + i)
     if world > 1:
         dist.barrier()
-    torch.cuda.synchronize()
+    _sync()
     elapsed = time.perf_counter() - t0
 
     # MAX over ranks
