@@ -16,7 +16,8 @@
 // Kernels:
 //   bn_stats<T>      per-channel Σx, Σx² (register accumulation, one
 //                    LDS tree + C global atomics per block)
-//   bn_finalize      mean/rstd + running-stat update (1 block)
+//   bn_reduce_finalize<FIN>  fused partial reduce + (fwd) mean/rstd +
+//                    running-stat update — one launch, no memset
 //   bn_fwd<T,RELU,ADD>   y = (x−μ)·rstd·γ + β (+z) (ReLU)
 //   bn_bwd_stats<T,RELU> per-channel Σdy_eff, Σdy_eff·x̂
 //   bn_bwd_dx<T,RELU,ADD> dx (+dz) from the reduced stats
@@ -124,18 +125,14 @@ std::vector<at::Tensor> bn_fwd_train(
         bn_stats_kernel<float><<<grid, BN_BLOCK, 0, stream>>>(
             x.data_ptr<float>(), m, c, partials.data_ptr<float>());
     }
-    {
-        const int slices = std::min(grid, std::max(1, 32768 / (2 * c)));
-        const long long work = (long long)2 * c * slices;
-        BN_OK(hipMemsetAsync(sums, 0, 2 * c * sizeof(float), stream));
-        bn_reduce_partials_kernel<<<(int)((work + BN_BLOCK - 1) / BN_BLOCK),
-                                    BN_BLOCK, 0, stream>>>(
-            partials.data_ptr<float>(), grid, c, slices, sums);
-    }
-    bn_finalize_kernel<<<(c + BN_BLOCK - 1) / BN_BLOCK, BN_BLOCK, 0,
-                         stream>>>(
-        sums, m, c, (float)eps, (float)momentum,
-        saved_mean.data_ptr<float>(), saved_rstd.data_ptr<float>(),
+    // fused reduce+finalize: one launch, no sums memset (writes every
+    // sums/saved/running entry it owns).  Grid hoisted into a variable:
+    // hipify mis-parses a `<` inside the launch config.
+    const int rf_grid = bn_rf_grid(c);
+    bn_reduce_finalize_kernel<true><<<rf_grid, BN_BLOCK, 0, stream>>>(
+        partials.data_ptr<float>(), grid, c, m, (float)eps,
+        (float)momentum, sums, saved_mean.data_ptr<float>(),
+        saved_rstd.data_ptr<float>(),
         running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
         running_var.defined() ? running_var.data_ptr<float>() : nullptr);
     at::Tensor mask;
@@ -218,13 +215,10 @@ std::vector<at::Tensor> bn_bwd(
             bn_bwd_stats_kernel<T, false><<<grid, BN_BLOCK, 0, stream>>>( \
                 dyp, xp, yp, m, c, saved_mean.data_ptr<float>(), \
                 saved_rstd.data_ptr<float>(), pp); \
-        { const int slices = std::min(grid, std::max(1, 32768 / (2 * c))); \
-          const long long work = (long long)2 * c * slices; \
-          BN_OK(hipMemsetAsync(sums, 0, 2 * c * sizeof(float), stream)); \
-          bn_reduce_partials_kernel<<<(int)((work + BN_BLOCK - 1) / \
-                                            BN_BLOCK), BN_BLOCK, 0, \
-                                      stream>>>(pp, grid, c, slices, \
-                                                sums); } \
+        bn_reduce_finalize_kernel<false> \
+            <<<bn_rf_grid(c), BN_BLOCK, 0, stream>>>( \
+            pp, grid, c, m, 0.0f, 0.0f, sums, nullptr, nullptr, \
+            nullptr, nullptr); \
         if (relu && need_dz) \
             bn_bwd_dx_kernel<T, true, true><<<grid, BN_BLOCK, 0, stream>>>( \
                 dyp, xp, yp, dxp, dzp, m, c, \

@@ -158,6 +158,71 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_reduce_partials_kernel(
     }
 }
 
+// ---- fused reduce(+finalize): ONE launch replacing memset + reduce +
+// finalize ----------------------------------------------------------------
+// The three-launch sequence above is launch-latency-bound at ~30 µs per
+// BN (memset 2C + reduce + finalize, each a tiny kernel); ×106 BN stats
+// ops per ResNet-50 step that is ~3 ms/step of glue.  One block owns
+// CH_PER=min(32,c) channels: threads form (BN_BLOCK/CH_PER) slices that
+// stride the partial rows (coalesced 128 B groups), reduce across
+// slices in LDS, then slice 0 writes sums[ch], sums[c+ch] and — in
+// FINALIZE mode (forward) — mean/rstd/running stats directly.  No
+// atomics, no zeroed scratch.  c is always a power of two here (C/W
+// pow2 is checked at dispatch), so CH_PER divides c and the slice
+// count is a power of two.
+inline int bn_rf_grid(int c) {  // host-side launch grid for the kernel
+    return c / (c < 32 ? c : 32);
+}
+
+template <bool FINALIZE>
+__global__ __launch_bounds__(BN_BLOCK) void bn_reduce_finalize_kernel(
+        const float* __restrict__ partials, int nblocks, int c,
+        long long m, float eps, float momentum,
+        float* __restrict__ sums, float* __restrict__ saved_mean,
+        float* __restrict__ saved_rstd, float* __restrict__ running_mean,
+        float* __restrict__ running_var) {
+    const int ch_per = c < 32 ? c : 32;
+    const int slices = BN_BLOCK / ch_per;
+    const int ch_off = threadIdx.x % ch_per;
+    const int slice = threadIdx.x / ch_per;
+    const int ch = blockIdx.x * ch_per + ch_off;
+    float s_acc = 0.0f, q_acc = 0.0f;
+    for (int b = slice; b < nblocks; b += slices) {
+        const float* row = partials + (long long)b * 2 * c;
+        s_acc += row[ch];
+        q_acc += row[c + ch];
+    }
+    __shared__ float red_s[BN_BLOCK];
+    __shared__ float red_q[BN_BLOCK];
+    red_s[threadIdx.x] = s_acc;
+    red_q[threadIdx.x] = q_acc;
+    __syncthreads();
+    for (int off = slices >> 1; off >= 1; off >>= 1) {
+        if (slice < off) {
+            red_s[threadIdx.x] += red_s[threadIdx.x + off * ch_per];
+            red_q[threadIdx.x] += red_q[threadIdx.x + off * ch_per];
+        }
+        __syncthreads();
+    }
+    if (slice != 0) return;
+    const float s = red_s[ch_off], q = red_q[ch_off];
+    sums[ch] = s;
+    sums[c + ch] = q;
+    if (FINALIZE) {
+        const float mean = s / (float)m;
+        float var = q / (float)m - mean * mean;
+        if (var < 0.0f) var = 0.0f;
+        saved_mean[ch] = mean;
+        saved_rstd[ch] = rsqrtf(var + eps);
+        if (running_mean != nullptr) {
+            const float unbiased = m > 1 ? var * (float)m / (float)(m - 1)
+                                         : var;
+            running_mean[ch] += momentum * (mean - running_mean[ch]);
+            running_var[ch] += momentum * (unbiased - running_var[ch]);
+        }
+    }
+}
+
 // ---- finalize: mean/rstd + running-stat update ----------------------------
 __global__ __launch_bounds__(BN_BLOCK) void bn_finalize_kernel(
         const float* __restrict__ sums, long long m, int c, float eps,
