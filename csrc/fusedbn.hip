@@ -125,12 +125,17 @@ std::vector<at::Tensor> bn_fwd_train(
         bn_stats_kernel<float><<<grid, BN_BLOCK, 0, stream>>>(
             x.data_ptr<float>(), m, c, partials.data_ptr<float>());
     }
-    // fused reduce+finalize: one launch, no sums memset (writes every
-    // sums/saved/running entry it owns).  Grid hoisted into a variable:
-    // hipify mis-parses a `<` inside the launch config.
+    // two-level reduce + fused finalize: no memsets, no atomics.
+    // Grids hoisted into variables: hipify mis-parses a `<` inside the
+    // launch config.
+    const int slices = bn_slices(grid, c);
+    const int l1_grid = (2 * c * slices + BN_BLOCK - 1) / BN_BLOCK;
     const int rf_grid = bn_rf_grid(c);
+    if (slices < grid)
+        bn_reduce_slices_kernel<<<l1_grid, BN_BLOCK, 0, stream>>>(
+            partials.data_ptr<float>(), grid, c, slices);
     bn_reduce_finalize_kernel<true><<<rf_grid, BN_BLOCK, 0, stream>>>(
-        partials.data_ptr<float>(), grid, c, m, (float)eps,
+        partials.data_ptr<float>(), slices, c, m, (float)eps,
         (float)momentum, sums, saved_mean.data_ptr<float>(),
         saved_rstd.data_ptr<float>(),
         running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
@@ -215,10 +220,15 @@ std::vector<at::Tensor> bn_bwd(
             bn_bwd_stats_kernel<T, false><<<grid, BN_BLOCK, 0, stream>>>( \
                 dyp, xp, yp, m, c, saved_mean.data_ptr<float>(), \
                 saved_rstd.data_ptr<float>(), pp); \
-        bn_reduce_finalize_kernel<false> \
-            <<<bn_rf_grid(c), BN_BLOCK, 0, stream>>>( \
-            pp, grid, c, m, 0.0f, 0.0f, sums, nullptr, nullptr, \
-            nullptr, nullptr); \
+        { const int slices = bn_slices(grid, c); \
+          const int l1_grid = (2 * c * slices + BN_BLOCK - 1) / BN_BLOCK; \
+          if (slices < grid) \
+              bn_reduce_slices_kernel<<<l1_grid, BN_BLOCK, 0, stream>>>( \
+                  pp, grid, c, slices); \
+          bn_reduce_finalize_kernel<false> \
+              <<<bn_rf_grid(c), BN_BLOCK, 0, stream>>>( \
+              pp, slices, c, m, 0.0f, 0.0f, sums, nullptr, nullptr, \
+              nullptr, nullptr); } \
         if (relu && need_dz) \
             bn_bwd_dx_kernel<T, true, true><<<grid, BN_BLOCK, 0, stream>>>( \
                 dyp, xp, yp, dxp, dzp, m, c, \

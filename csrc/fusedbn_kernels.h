@@ -138,38 +138,47 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_stats_kernel(
     }
 }
 
-// ---- reduce per-block partials into sums[2C] ------------------------------
+// ---- level-1 partial reduce: nblocks rows -> slices rows, IN PLACE -------
 // Parallel over (element, block-slice): 2C alone is far too few threads
 // (a 2C-thread serial loop over 2048 rows measured 474 µs — more than
-// the stats pass itself).  sums must be zeroed by the caller.
-__global__ __launch_bounds__(BN_BLOCK) void bn_reduce_partials_kernel(
-        const float* __restrict__ partials, int nblocks, int c,
-        int slices, float* __restrict__ sums) {
+// the stats pass itself).  Each (element i, slice s) thread sums rows
+// {s, s+slices, …} of column i and writes the result back to
+// partials[s][i]: that cell's only reader is this same thread (row s
+// belongs to slice s), and its first read IS row s, so the in-place
+// write is race-free and deterministic — no atomics, no zeroed
+// accumulator, no extra scratch.
+__global__ __launch_bounds__(BN_BLOCK) void bn_reduce_slices_kernel(
+        float* __restrict__ partials, int nblocks, int c, int slices) {
     const long long idx = (long long)blockIdx.x * BN_BLOCK + threadIdx.x;
     const long long total = (long long)2 * c * slices;
-    for (long long t = idx; t < total;
-         t += (long long)gridDim.x * BN_BLOCK) {
-        const int i = (int)(t % (2 * c));   // consecutive threads: coalesced
-        const int s = (int)(t / (2 * c));
-        float acc = 0.0f;
-        for (int b = s; b < nblocks; b += slices)
-            acc += partials[(long long)b * 2 * c + i];
-        atomicAdd(&sums[i], acc);
-    }
+    if (idx >= total) return;
+    const int i = (int)(idx % (2 * c));     // consecutive threads: coalesced
+    const int s = (int)(idx / (2 * c));
+    float acc = 0.0f;
+    for (int b = s; b < nblocks; b += slices)
+        acc += partials[(long long)b * 2 * c + i];
+    partials[(long long)s * 2 * c + i] = acc;
 }
 
-// ---- fused reduce(+finalize): ONE launch replacing memset + reduce +
-// finalize ----------------------------------------------------------------
-// The three-launch sequence above is launch-latency-bound at ~30 µs per
-// BN (memset 2C + reduce + finalize, each a tiny kernel); ×106 BN stats
-// ops per ResNet-50 step that is ~3 ms/step of glue.  One block owns
-// CH_PER=min(32,c) channels: threads form (BN_BLOCK/CH_PER) slices that
-// stride the partial rows (coalesced 128 B groups), reduce across
-// slices in LDS, then slice 0 writes sums[ch], sums[c+ch] and — in
-// FINALIZE mode (forward) — mean/rstd/running stats directly.  No
-// atomics, no zeroed scratch.  c is always a power of two here (C/W
-// pow2 is checked at dispatch), so CH_PER divides c and the slice
-// count is a power of two.
+inline int bn_slices(int nblocks, int c) {
+    const int s = 32768 / (2 * c) > 1 ? 32768 / (2 * c) : 1;
+    return nblocks < s ? nblocks : s;
+}
+
+// ---- level-2 fused reduce + finalize -------------------------------------
+// Replaces the old memset(2C) + atomicAdd reduce + finalize launch trio
+// (launch-latency-bound at ~30 µs per BN; ×106 BN stats ops per
+// ResNet-50 step ≈ 3 ms/step of glue).  Reduces the ≤ bn_slices() rows
+// left by bn_reduce_slices_kernel.  One block owns CH_PER=min(32,c)
+// channels: threads form (BN_BLOCK/CH_PER) row-slices (coalesced 128 B
+// groups), reduce across them in LDS, then slice 0 writes sums[ch],
+// sums[c+ch] and — in FINALIZE mode (forward) — mean/rstd/running
+// stats directly.  No atomics, no zeroed scratch.  Parallelism is ∝ c,
+// which is why level 1 must first shrink the row count: at c=64 this
+// kernel is 2 blocks, and looping it over 2048 raw partial rows was
+// measured 6 ms/step SLOWER than the trio it replaced.  c is always a
+// power of two here (C/W pow2 is checked at dispatch), so CH_PER
+// divides c and the slice count is a power of two.
 inline int bn_rf_grid(int c) {  // host-side launch grid for the kernel
     return c / (c < 32 ? c : 32);
 }
@@ -219,28 +228,6 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_reduce_finalize_kernel(
                                          : var;
             running_mean[ch] += momentum * (mean - running_mean[ch]);
             running_var[ch] += momentum * (unbiased - running_var[ch]);
-        }
-    }
-}
-
-// ---- finalize: mean/rstd + running-stat update ----------------------------
-__global__ __launch_bounds__(BN_BLOCK) void bn_finalize_kernel(
-        const float* __restrict__ sums, long long m, int c, float eps,
-        float momentum, float* __restrict__ saved_mean,
-        float* __restrict__ saved_rstd, float* __restrict__ running_mean,
-        float* __restrict__ running_var) {
-    for (int i = blockIdx.x * BN_BLOCK + threadIdx.x; i < c;
-         i += gridDim.x * BN_BLOCK) {
-        const float mean = sums[i] / (float)m;
-        float var = sums[c + i] / (float)m - mean * mean;
-        if (var < 0.0f) var = 0.0f;
-        saved_mean[i] = mean;
-        saved_rstd[i] = rsqrtf(var + eps);
-        if (running_mean != nullptr) {
-            const float unbiased = m > 1 ? var * (float)m / (float)(m - 1)
-                                         : var;
-            running_mean[i] += momentum * (mean - running_mean[i]);
-            running_var[i] += momentum * (unbiased - running_var[i]);
         }
     }
 }
