@@ -161,7 +161,13 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_reduce_slices_kernel(
 }
 
 inline int bn_slices(int nblocks, int c) {
-    const int s = 32768 / (2 * c) > 1 ? 32768 / (2 * c) : 1;
+    // Balance the two levels' per-thread loop lengths: level 2 runs
+    // BN_BLOCK/CH_PER row-slices per block, so slices = 16× that gives
+    // it ≤16 iterations, and level 1 then does nblocks/slices ≤ 16
+    // iterations at the 2048-block grid cap.  (The old 32768-thread
+    // cap was sized for atomicAdd chain length — atomics are gone.)
+    const int ch_per = c < 32 ? c : 32;
+    const int s = 16 * (BN_BLOCK / ch_per);
     return nblocks < s ? nblocks : s;
 }
 
