@@ -68,7 +68,9 @@ template <> struct VecTraits<float> {
 template <typename T>
 __global__ __launch_bounds__(BN_BLOCK) void bn_stats_kernel(
         const T* __restrict__ x, long long m, int c,
-        float* __restrict__ partials /* [gridDim][2C], zero-init */) {
+        float* __restrict__ partials /* [gridDim][2C]; fully overwritten
+                                        when ch_chunks <= BLOCK, else
+                                        caller zero-inits */) {
     constexpr int W = VecTraits<T>::W;
     const int ch_chunks = c / W;
     float s[W] = {0}, q[W] = {0};
@@ -301,7 +303,9 @@ __global__ __launch_bounds__(BN_BLOCK) void bn_bwd_stats_kernel(
         const T* __restrict__ dy, const T* __restrict__ x,
         const unsigned char* __restrict__ mask, long long m, int c,
         const float* __restrict__ mean, const float* __restrict__ rstd,
-        float* __restrict__ partials /* [gridDim][2C], zero-init */) {
+        float* __restrict__ partials /* [gridDim][2C]; fully overwritten
+                                        when ch_chunks <= BLOCK, else
+                                        caller zero-inits */) {
     constexpr int W = VecTraits<T>::W;
     const int ch_chunks = c / W;
     long long r0, r_limit, rstep;
