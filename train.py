@@ -30,7 +30,8 @@ def parse_args():
     p = argparse.ArgumentParser(description="gTop-k S-SGD (MI355X-native)")
     p.add_argument("--dnn", default="resnet20")
     p.add_argument("--dataset", default="cifar10",
-                   choices=["mnist", "cifar10", "imagenet", "ptb", "an4"])
+                   choices=["mnist", "cifar10", "imagenet", "ptb", "an4",
+                            "an4wav"])
     p.add_argument("--data-dir", default="./data", help="unused (synthetic)")
     p.add_argument("--batch-size", type=int, default=32)
     p.add_argument("--nsteps-update", type=int, default=1)
