@@ -41,6 +41,32 @@ def test_train_cli_gloo_world2(tmp_path):
     assert r.returncode == 0, (r.stdout[-1000:], r.stderr[-2000:])
 
 
+def _torchrun_train(nproc: int, port: int, tmp_path, timeout=420):
+    env = dict(os.environ)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", str(nproc), "--master-addr", "127.0.0.1",
+         "--master-port", str(port), os.path.join(REPO, "train.py"),
+         "--dnn", "mnistffn", "--dataset", "mnist", "--batch-size", "8",
+         "--max-epochs", "1", "--iters-per-epoch", "2", "--n-samples", "64",
+         "--density", "0.05", "--device", "cpu", "--no-amp",
+         "--num-workers", "0",
+         "--checkpoint-dir", str(tmp_path / "ckpt")],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=timeout)
+    assert r.returncode == 0, (r.stdout[-1000:], r.stderr[-2000:])
+
+
+def test_train_cli_gloo_world3(tmp_path):
+    """Non-power-of-2 world: the halving merge's peer-existence edge
+    (rank 2 pairs with a missing rank 3) end-to-end through torchrun."""
+    _torchrun_train(3, 29518, tmp_path)
+
+
+def test_train_cli_gloo_world4(tmp_path):
+    """Two full merge rounds (2/4/8-proc single-node)."""
+    _torchrun_train(4, 29519, tmp_path)
+
+
 def test_collective_self_benchmark_cpu():
     r = subprocess.run(
         [sys.executable, "-m", "gtopkssgd_amd.parallel.allreducer",
